@@ -1,0 +1,208 @@
+/* ORACLE (test infrastructure only).
+ *
+ * validate_populated_transaction_and_get_fee + check_scripts + muhash reduce,
+ * restating consensus/src/processes/transaction_validator/
+ * tx_validation_in_utxo_context.rs:37-218 and consensus/src/pipeline/
+ * virtual_processor/utxo_validation.rs:297-348.
+ *
+ * Round-1 scope notes (documented in DESIGN.md):
+ *  - KV_FLAGS_FULL's storage-mass commitment check (check_mass_commitment,
+ *    :126-134) is not yet restated; callers must pass KV_FLAGS_SKIP_MASS_CHECK
+ *    (the flag the reference's own validator tests use). Full returns
+ *    KV_ERR_BAD_BLOB to fail loudly rather than silently skip.
+ *  - Covenant context (check_covenant_info) is out of scope; blobs carrying
+ *    covenant bindings are rejected the same way.
+ */
+#include "kaspa_engine_abi.h"
+#include "ok_tx.h"
+#include <string.h>
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+#define MAX_SOMPI (29000000000ULL * 100000000ULL)
+#define SEQ_DISABLED (1ULL << 63)
+#define SEQ_MASK 0x00000000ffffffffULL
+
+/* default consensus params (config/params.rs:582-650 mainnet) */
+#define DEFAULT_COINBASE_MATURITY 1000
+#define DEFAULT_MASS_PER_SIG_OP 1000
+
+static int validate_one_tx(const ok_tx *tx, uint64_t pov_daa, uint32_t flags,
+                           uint64_t coinbase_maturity, uint64_t mass_per_sig_op,
+                           uint64_t *fee_out) {
+  /* 1. coinbase maturity (:81-97) */
+  for (uint32_t i = 0; i < tx->n_inputs; i++) {
+    const ok_input *in = &tx->inputs[i];
+    if (in->utxo_is_coinbase && in->utxo_daa_score + coinbase_maturity > pov_daa)
+      return KV_ERR_IMMATURE_COINBASE;
+  }
+  /* 2. input amounts (:99-114) */
+  uint64_t total_in = 0;
+  for (uint32_t i = 0; i < tx->n_inputs; i++) {
+    uint64_t amt = tx->inputs[i].utxo_amount;
+    if (total_in + amt < total_in) return KV_ERR_INPUT_AMOUNT_OVERFLOW;
+    total_in += amt;
+    if (total_in > MAX_SOMPI) return KV_ERR_INPUT_AMOUNT_TOO_HIGH;
+  }
+  /* 3. outputs (:116-124) */
+  uint64_t total_out = 0;
+  for (uint32_t i = 0; i < tx->n_outputs; i++) total_out += tx->outputs[i].value;
+  if (total_in < total_out) return KV_ERR_SPEND_TOO_HIGH;
+  uint64_t fee = total_in - total_out;
+  /* 4. mass commitment — round-1: must be skipped explicitly */
+  if (flags != KV_FLAGS_SKIP_MASS_CHECK && flags != KV_FLAGS_SKIP_SCRIPT_CHECKS)
+    return KV_ERR_BAD_BLOB;
+  /* 5. sequence locks (:136-161) */
+  for (uint32_t i = 0; i < tx->n_inputs; i++) {
+    const ok_input *in = &tx->inputs[i];
+    if ((in->sequence & SEQ_DISABLED) == SEQ_DISABLED) continue;
+    int64_t relative_lock = (int64_t)(in->sequence & SEQ_MASK);
+    int64_t lock_daa = (int64_t)in->utxo_daa_score + relative_lock - 1;
+    if (lock_daa >= (int64_t)pov_daa) return KV_ERR_SEQUENCE_LOCK;
+  }
+  /* 6. covenants: reject blobs with covenant data (round-1 out of scope) */
+  for (uint32_t i = 0; i < tx->n_outputs; i++)
+    if (tx->outputs[i].has_covenant) return KV_ERR_BAD_BLOB;
+  /* 7. scripts (:60-65,163-218) */
+  if (flags != KV_FLAGS_SKIP_SCRIPT_CHECKS) {
+    ok_sighash_reused reused;
+    ok_reused_init(&reused);
+    for (uint32_t i = 0; i < tx->n_inputs; i++) {
+      int src = ok_script_check_input(tx, i, mass_per_sig_op, &reused);
+      if (src) {
+        /* map_script_err (:220-222) */
+        return tx->inputs[i].sig_script_len == 0 ? KV_ERR_SIGNATURE_EMPTY_BASE + src
+                                                 : KV_ERR_SIGNATURE_INVALID_BASE + src;
+      }
+    }
+  }
+  *fee_out = fee;
+  return KV_OK;
+}
+
+static int validate_impl(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_score,
+                         uint64_t block_daa_score, uint32_t flags, int threads,
+                         int32_t *tx_codes_out, uint64_t *fees_out,
+                         uint8_t muhash_out[32]) {
+  if (blob_len < 4) return -1;
+  uint32_t n_txs;
+  memcpy(&n_txs, blob, 4);
+
+  uint64_t num[OK_U3072_LIMBS], den[OK_U3072_LIMBS];
+  ok_u3072_one(num);
+  ok_u3072_one(den);
+  int bad = 0;
+
+#ifdef _OPENMP
+  if (threads > 1) {
+    omp_set_num_threads(threads);
+#pragma omp parallel
+    {
+      uint64_t lnum[OK_U3072_LIMBS], lden[OK_U3072_LIMBS];
+      ok_u3072_one(lnum);
+      ok_u3072_one(lden);
+#pragma omp for schedule(dynamic, 4)
+      for (uint32_t t = 0; t < n_txs; t++) {
+        ok_tx tx;
+        if (ok_tx_parse(blob, blob_len, t, &tx)) {
+          tx_codes_out[t] = KV_ERR_BAD_BLOB;
+          fees_out[t] = 0;
+          __atomic_store_n(&bad, 1, __ATOMIC_RELAXED);
+          continue;
+        }
+        uint64_t fee = 0;
+        int code = validate_one_tx(&tx, pov_daa_score, flags, DEFAULT_COINBASE_MATURITY,
+                                   DEFAULT_MASS_PER_SIG_OP, &fee);
+        tx_codes_out[t] = code;
+        fees_out[t] = fee;
+        if (code == KV_OK && muhash_out)
+          ok_tx_muhash(&tx, block_daa_score, lnum, lden);
+        ok_tx_free(&tx);
+      }
+      if (muhash_out) {
+#pragma omp critical
+        {
+          ok_u3072_mul(num, lnum);
+          ok_u3072_mul(den, lden);
+        }
+      }
+    }
+    if (muhash_out) ok_muhash_finalize(num, den, muhash_out);
+    return bad ? -1 : 0;
+  }
+#endif
+  (void)threads;
+  for (uint32_t t = 0; t < n_txs; t++) {
+    ok_tx tx;
+    if (ok_tx_parse(blob, blob_len, t, &tx)) {
+      tx_codes_out[t] = KV_ERR_BAD_BLOB;
+      fees_out[t] = 0;
+      bad = 1;
+      continue;
+    }
+    uint64_t fee = 0;
+    int code = validate_one_tx(&tx, pov_daa_score, flags, DEFAULT_COINBASE_MATURITY,
+                               DEFAULT_MASS_PER_SIG_OP, &fee);
+    tx_codes_out[t] = code;
+    fees_out[t] = fee;
+    if (code == KV_OK && muhash_out) ok_tx_muhash(&tx, block_daa_score, num, den);
+    ok_tx_free(&tx);
+  }
+  if (muhash_out) ok_muhash_finalize(num, den, muhash_out);
+  return bad ? -1 : 0;
+}
+
+int ok_validate_block(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_score,
+                      uint64_t block_daa_score, uint32_t flags, int32_t *tx_codes_out,
+                      uint64_t *fees_out, uint8_t muhash_out[32]) {
+  return validate_impl(blob, blob_len, pov_daa_score, block_daa_score, flags, 1,
+                       tx_codes_out, fees_out, muhash_out);
+}
+
+int ok_validate_block_parallel(const uint8_t *blob, size_t blob_len,
+                               uint64_t pov_daa_score, uint64_t block_daa_score,
+                               uint32_t flags, int threads, int32_t *tx_codes_out,
+                               uint64_t *fees_out, uint8_t muhash_out[32]) {
+  return validate_impl(blob, blob_len, pov_daa_score, block_daa_score, flags, threads,
+                       tx_codes_out, fees_out, muhash_out);
+}
+
+int ok_check_input_script(const uint8_t *blob, size_t blob_len, uint32_t tx_index,
+                          uint32_t input_index) {
+  ok_tx tx;
+  if (ok_tx_parse(blob, blob_len, tx_index, &tx)) return -1;
+  if (input_index >= tx.n_inputs) {
+    ok_tx_free(&tx);
+    return -1;
+  }
+  ok_sighash_reused reused;
+  ok_reused_init(&reused);
+  int rc = ok_script_check_input(&tx, input_index, DEFAULT_MASS_PER_SIG_OP, &reused);
+  ok_tx_free(&tx);
+  return rc;
+}
+
+/* ---------------- parallel schnorr-tuple verify (CPU baseline for config 2) ----
+ * tuples: n × 128B = r‖s‖pk‖msg (the kv_verify_schnorr_batch layout). */
+int ok_verify_schnorr_batch(const uint8_t *tuples, size_t n, int threads,
+                            uint64_t *bitmap_out) {
+  size_t words = (n + 63) / 64;
+  memset(bitmap_out, 0, words * 8);
+#ifdef _OPENMP
+  omp_set_num_threads(threads > 0 ? threads : 1);
+#pragma omp parallel for schedule(dynamic, 64)
+#endif
+  for (size_t i = 0; i < n; i++) {
+    const uint8_t *t = tuples + i * 128;
+    uint8_t sig[64];
+    memcpy(sig, t, 64); /* r ‖ s */
+    int v = ok_schnorr_verify(t + 64, t + 96, sig) == 1;
+    if (v)
+#ifdef _OPENMP
+#pragma omp atomic
+#endif
+      bitmap_out[i / 64] |= (1ULL << (i % 64));
+  }
+  return 0;
+}

@@ -120,8 +120,15 @@ int ok_validate_block_parallel(const uint8_t *blob, size_t blob_len, uint64_t po
 int ok_sighash(const uint8_t *blob, size_t blob_len, uint32_t tx_index, uint32_t input_index,
                uint8_t hash_type, int ecdsa, uint8_t out32[32]);
 
+/* transaction hash (hashing/tx.rs:20-24) */
+int ok_tx_hash_blob(const uint8_t *blob, size_t blob_len, uint32_t tx_index, uint8_t out32[32]);
+
 /* transaction id (consensus/core/src/hashing/tx.rs:34-48,207-218) */
 int ok_tx_id(const uint8_t *blob, size_t blob_len, uint32_t tx_index, uint8_t out32[32]);
+
+/* CPU-baseline batched schnorr verify over n × 128B (r‖s‖pk‖msg) tuples */
+int ok_verify_schnorr_batch(const uint8_t *tuples, size_t n, int threads,
+                            uint64_t *bitmap_out);
 
 /* run the script engine for one input; returns 0 ok, else KV script error code */
 int ok_check_input_script(const uint8_t *blob, size_t blob_len, uint32_t tx_index,

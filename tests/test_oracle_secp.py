@@ -1,0 +1,136 @@
+"""Oracle secp256k1 self-tests: BIP-340 vector, sign→verify round trips, tamper
+rejection, high-S/overflow handling, and a cross-check against an independent
+pure-python bigint EC implementation. (The mainnet-signature parity lives in
+test_oracle_script.py — those pin us to the reference's libsecp256k1.)"""
+import ctypes
+import hashlib
+import random
+
+P = 0xFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFEFFFFFC2F
+N = 0xFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFFEBAAEDCE6AF48A03BBFD25E8CD0364141
+G = (0x79BE667EF9DCBBAC55A06295CE870B07029BFCDB2DCE28D959F2815B16F81798,
+     0x483ADA7726A3C4655DA4FBFC0E1108A8FD17B448A68554199C47D08FFB10D4B8)
+
+
+def _add(Pt, Q):
+    if Pt is None:
+        return Q
+    if Q is None:
+        return Pt
+    if Pt[0] == Q[0] and (Pt[1] + Q[1]) % P == 0:
+        return None
+    if Pt == Q:
+        lam = 3 * Pt[0] * Pt[0] * pow(2 * Pt[1], P - 2, P) % P
+    else:
+        lam = (Q[1] - Pt[1]) * pow(Q[0] - Pt[0], P - 2, P) % P
+    x = (lam * lam - Pt[0] - Q[0]) % P
+    return (x, (lam * (Pt[0] - x) - Pt[1]) % P)
+
+
+def _mul(k, Pt):
+    R = None
+    while k:
+        if k & 1:
+            R = _add(R, Pt)
+        Pt = _add(Pt, Pt)
+        k >>= 1
+    return R
+
+
+def _th(tag, data):
+    t = hashlib.sha256(tag.encode()).digest()
+    return hashlib.sha256(t + t + data).digest()
+
+
+def _pysign(sk, msg):
+    Pt = _mul(sk, G)
+    d = sk if Pt[1] % 2 == 0 else N - sk
+    pkb = Pt[0].to_bytes(32, "big")
+    t = (d ^ int.from_bytes(_th("BIP0340/aux", bytes(32)), "big")).to_bytes(32, "big")
+    k0 = int.from_bytes(_th("BIP0340/nonce", t + pkb + msg), "big") % N
+    R = _mul(k0, G)
+    k = k0 if R[1] % 2 == 0 else N - k0
+    rb = R[0].to_bytes(32, "big")
+    e = int.from_bytes(_th("BIP0340/challenge", rb + pkb + msg), "big") % N
+    return pkb, rb + ((k + e * d) % N).to_bytes(32, "big")
+
+
+def test_bip340_vector(oracle):
+    pk = (ctypes.c_uint8 * 32)()
+    sig = (ctypes.c_uint8 * 64)()
+    sk3 = (3).to_bytes(32, "big")
+    assert oracle.ok_pubkey_xonly(sk3, pk) == 1
+    assert bytes(pk).hex() == "f9308a019258c31049344f85f89d5229b531c845836f99b08601f113bce036f9"
+    assert oracle.ok_schnorr_sign(sk3, bytes(32), bytes(32), sig) == 1
+    # expected value independently recomputed with pure-python EC (matches _pysign)
+    _, exp = _pysign(3, bytes(32))
+    assert bytes(sig) == exp
+    assert oracle.ok_schnorr_verify(pk, bytes(32), sig) == 1
+
+
+def test_schnorr_roundtrip_and_tamper(oracle):
+    rng = random.Random(42)
+    pk = (ctypes.c_uint8 * 32)()
+    sig = (ctypes.c_uint8 * 64)()
+    for i in range(25):
+        sk = rng.getrandbits(256).to_bytes(32, "big")
+        m = rng.getrandbits(256).to_bytes(32, "big")
+        if not oracle.ok_pubkey_xonly(sk, pk):
+            continue
+        assert oracle.ok_schnorr_sign(sk, m, rng.getrandbits(256).to_bytes(32, "big"), sig) == 1
+        assert oracle.ok_schnorr_verify(pk, m, sig) == 1
+        bad = bytearray(bytes(sig))
+        bad[rng.randrange(64)] ^= 1
+        assert oracle.ok_schnorr_verify(pk, m, bytes(bad)) == 0
+        badm = bytearray(m)
+        badm[0] ^= 1
+        assert oracle.ok_schnorr_verify(pk, bytes(badm), sig) == 0
+
+
+def test_cross_implementation(oracle):
+    """Signatures from the independent python signer must verify in C."""
+    rng = random.Random(1)
+    for _ in range(6):
+        sk = rng.getrandbits(255) | 1
+        m = rng.getrandbits(256).to_bytes(32, "big")
+        pkb, sig = _pysign(sk, m)
+        assert oracle.ok_schnorr_verify(pkb, m, sig) == 1
+
+
+def test_ecdsa_roundtrip_highs_overflow(oracle):
+    rng = random.Random(9)
+    pk33 = (ctypes.c_uint8 * 33)()
+    sig = (ctypes.c_uint8 * 64)()
+    for i in range(25):
+        sk = rng.getrandbits(256).to_bytes(32, "big")
+        m = rng.getrandbits(256).to_bytes(32, "big")
+        if not oracle.ok_pubkey_compressed(sk, pk33):
+            continue
+        assert oracle.ok_ecdsa_sign(sk, m, sig) == 1
+        assert oracle.ok_ecdsa_verify(pk33, m, sig) == 1
+        s = int.from_bytes(bytes(sig)[32:], "big")
+        high_s = bytes(sig)[:32] + (N - s).to_bytes(32, "big")
+        assert oracle.ok_ecdsa_verify(pk33, m, high_s) == 0  # libsecp rejects high-S
+        overflow = (N + 1).to_bytes(32, "big") + bytes(sig)[32:]
+        assert oracle.ok_ecdsa_verify(pk33, m, overflow) == -2  # parse overflow
+
+    # invalid pubkey prefix
+    bad_pk = bytes([0x04]) + bytes(pk33)[1:]
+    assert oracle.ok_ecdsa_verify(bad_pk, m, sig) == -1
+
+
+def test_invalid_xonly_pubkey_is_parse_error(oracle):
+    """x with no curve point → InvalidPubkey (error, not boolean false)."""
+    sig = bytes(64)
+    msg = bytes(32)
+    found = 0
+    for x in range(2, 50):
+        r = oracle.ok_schnorr_verify(x.to_bytes(32, "big"), msg, sig)
+        rhs = (x**3 + 7) % P
+        on_curve = pow(rhs, (P - 1) // 2, P) == 1
+        if on_curve:
+            assert r in (0, 1)
+        else:
+            assert r == -1
+            found += 1
+    assert found > 0
