@@ -183,6 +183,20 @@ int kv_muhash_finalize(kv_ctx *ctx, const uint8_t *partial768, uint8_t *hash32_o
  * multiplied mod 2^3072-1103717) — MuHash::combine (crypto/muhash/src/lib.rs:87). */
 int kv_muhash_combine(kv_ctx *ctx, uint8_t *acc_partial768, const uint8_t *other_partial768);
 
+/* Batched verifies with per-signature status bytes (0 valid / 1 invalid /
+ * 2 pubkey-parse-error / 3 sig-parse-error) — the validate path uses these to
+ * distinguish TxScriptError::InvalidPubkey/InvalidSignature from boolean false. */
+int kv_verify_schnorr_batch_status(kv_ctx *ctx, const uint8_t *tuples, size_t n,
+                                   uint64_t *bitmap_out, uint8_t *status_out);
+int kv_verify_ecdsa_batch_status(kv_ctx *ctx, const uint8_t *tuples, size_t n,
+                                 uint64_t *bitmap_out, uint8_t *status_out);
+
+/* Staged mode: stage a batch into HBM once, then time repeated launches with
+ * inputs resident (bench harness; kernel_ms from hipEvents on the engine stream). */
+int kv_stage_tuples(kv_ctx *ctx, const uint8_t *tuples, size_t n, int ecdsa);
+int kv_verify_staged(kv_ctx *ctx, size_t n, int ecdsa, double *kernel_ms);
+int kv_fetch_bitmap(kv_ctx *ctx, size_t n, uint64_t *bitmap_out);
+
 /* Sig-cache statistics (crypto/txscript/src/caches.rs:57-82 counters). */
 typedef struct {
   uint64_t insertions;

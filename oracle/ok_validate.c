@@ -206,3 +206,68 @@ int ok_verify_schnorr_batch(const uint8_t *tuples, size_t n, int threads,
   }
   return 0;
 }
+
+/* ---------------- synthetic workload generation (bench/test harness only) ----
+ * Deterministic (pk,msg,sig) tuples for BASELINE config 2: keys/messages derived
+ * from a seed via keyed blake2b, signatures real BIP-340 (aux = zeros).
+ * invalid_permille of tuples get one sig byte flipped. */
+int ok_gen_schnorr_tuples(uint64_t seed, size_t n, uint32_t invalid_permille,
+                          uint8_t *out /* n × 128: r‖s‖pk‖msg */, int threads) {
+#ifdef _OPENMP
+  omp_set_num_threads(threads > 0 ? threads : 1);
+#pragma omp parallel for schedule(dynamic, 64)
+#endif
+  for (size_t i = 0; i < n; i++) {
+    uint8_t buf[16], sk[32], pk[32], msg[32], sig[64];
+    memcpy(buf, &seed, 8);
+    uint64_t idx = i;
+    memcpy(buf + 8, &idx, 8);
+    for (int attempt = 0;; attempt++) {
+      uint8_t tag = (uint8_t)attempt;
+      uint8_t b2[17];
+      memcpy(b2, buf, 16);
+      b2[16] = tag;
+      ok_blake2b_keyed((const uint8_t *)"kv-gen-key", 10, b2, 17, sk);
+      if (ok_pubkey_xonly(sk, pk)) break;
+    }
+    ok_blake2b_keyed((const uint8_t *)"kv-gen-msg", 10, buf, 16, msg);
+    ok_schnorr_sign(sk, msg, NULL, sig);
+    if (invalid_permille && (i % 1000) < invalid_permille) sig[40] ^= 0x20;
+    uint8_t *t = out + i * 128;
+    memcpy(t, sig, 64);
+    memcpy(t + 64, pk, 32);
+    memcpy(t + 96, msg, 32);
+  }
+  return 0;
+}
+
+/* ECDSA tuples: n × 132: r‖s‖pk33‖msg‖pad3 */
+int ok_gen_ecdsa_tuples(uint64_t seed, size_t n, uint32_t invalid_permille,
+                        uint8_t *out, int threads) {
+#ifdef _OPENMP
+  omp_set_num_threads(threads > 0 ? threads : 1);
+#pragma omp parallel for schedule(dynamic, 64)
+#endif
+  for (size_t i = 0; i < n; i++) {
+    uint8_t buf[16], sk[32], pk[33], msg[32], sig[64];
+    memcpy(buf, &seed, 8);
+    uint64_t idx = i;
+    memcpy(buf + 8, &idx, 8);
+    for (int attempt = 0;; attempt++) {
+      uint8_t b2[17];
+      memcpy(b2, buf, 16);
+      b2[16] = (uint8_t)attempt;
+      ok_blake2b_keyed((const uint8_t *)"kv-gen-ekey", 11, b2, 17, sk);
+      if (ok_pubkey_compressed(sk, pk)) break;
+    }
+    ok_blake2b_keyed((const uint8_t *)"kv-gen-emsg", 11, buf, 16, msg);
+    ok_ecdsa_sign(sk, msg, sig);
+    if (invalid_permille && (i % 1000) < invalid_permille) sig[40] ^= 0x20;
+    uint8_t *t = out + i * 132;
+    memcpy(t, sig, 64);
+    memcpy(t + 64, pk, 33);
+    memcpy(t + 97, msg, 32);
+    t[129] = t[130] = t[131] = 0;
+  }
+  return 0;
+}

@@ -126,6 +126,10 @@ int ok_tx_hash_blob(const uint8_t *blob, size_t blob_len, uint32_t tx_index, uin
 /* transaction id (consensus/core/src/hashing/tx.rs:34-48,207-218) */
 int ok_tx_id(const uint8_t *blob, size_t blob_len, uint32_t tx_index, uint8_t out32[32]);
 
+/* synthetic workload generation (harness only; deterministic by seed) */
+int ok_gen_schnorr_tuples(uint64_t seed, size_t n, uint32_t invalid_permille, uint8_t *out, int threads);
+int ok_gen_ecdsa_tuples(uint64_t seed, size_t n, uint32_t invalid_permille, uint8_t *out, int threads);
+
 /* CPU-baseline batched schnorr verify over n × 128B (r‖s‖pk‖msg) tuples */
 int ok_verify_schnorr_batch(const uint8_t *tuples, size_t n, int threads,
                             uint64_t *bitmap_out);

@@ -1,0 +1,493 @@
+/* MI355X-native secp256k1 device arithmetic (gfx950).
+ *
+ * PRODUCT code — the compute path of the engine. Written for the CDNA4
+ * execution model: one signature per lane, 4×u64 limb field elements held in
+ * VGPRs, carry chains expressed through __umul64hi / add-with-carry patterns
+ * the AMDGPU backend lowers to v_mad_u64_u32 / v_add*_co chains, branch-free
+ * table selection (no per-lane divergent memory indexing into scratch).
+ *
+ * Replaces the verification math of the vendored libsecp256k1 used by
+ * crypto/txscript/src/lib.rs:869 (schnorr) and :899 (ecdsa). Independent
+ * implementation from oracle/ok_secp.c (different structure: branchless,
+ * fused reduction, Jacobian-only checks without affine conversion where
+ * possible); parity is established by tests against the oracle and the
+ * reference's mainnet-signature vectors.
+ */
+#ifndef KV_SECP_DEVICE_H
+#define KV_SECP_DEVICE_H
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+namespace kv {
+
+typedef uint64_t u64;
+typedef uint32_t u32;
+
+/* ---------- 256-bit field element mod p = 2^256 - 0x1000003D1 ---------- */
+
+struct fe {
+  u64 n[4];
+};
+
+#define KV_P0 0xFFFFFFFEFFFFFC2FULL
+#define KV_P1 0xFFFFFFFFFFFFFFFFULL
+#define KV_PC 0x1000003D1ULL /* 2^256 - p */
+
+__device__ __forceinline__ u64 addc(u64 a, u64 b, u64 &carry) {
+  u64 s = a + b;
+  u64 c1 = s < a;
+  u64 s2 = s + carry;
+  carry = c1 + (s2 < s);
+  return s2;
+}
+
+__device__ __forceinline__ u64 subb(u64 a, u64 b, u64 &borrow) {
+  u64 d = a - b;
+  u64 b1 = a < b;
+  u64 d2 = d - borrow;
+  borrow = b1 + (d < borrow);
+  return d2;
+}
+
+__device__ __forceinline__ int fe_gte_p(const fe &a) {
+  /* a >= p ? branchless */
+  if (a.n[3] != KV_P1 || a.n[2] != KV_P1 || a.n[1] != KV_P1) {
+    return (a.n[3] == KV_P1 && a.n[2] == KV_P1 && a.n[1] == KV_P1) ? (a.n[0] >= KV_P0)
+                                                                   : 0;
+  }
+  return a.n[0] >= KV_P0;
+}
+
+__device__ __forceinline__ void fe_sub_p(fe &a) {
+  u64 borrow = 0;
+  a.n[0] = subb(a.n[0], KV_P0, borrow);
+  a.n[1] = subb(a.n[1], KV_P1, borrow);
+  a.n[2] = subb(a.n[2], KV_P1, borrow);
+  a.n[3] = subb(a.n[3], KV_P1, borrow);
+}
+
+__device__ __forceinline__ void fe_norm_once(fe &a) {
+  /* conditional subtract p, branchless */
+  u64 ge = (a.n[3] == KV_P1) & (a.n[2] == KV_P1) & (a.n[1] == KV_P1) &
+           (a.n[0] >= KV_P0);
+  u64 mask = 0 - ge;
+  u64 borrow = 0;
+  a.n[0] = subb(a.n[0], KV_P0 & mask, borrow);
+  a.n[1] = subb(a.n[1], KV_P1 & mask, borrow);
+  a.n[2] = subb(a.n[2], KV_P1 & mask, borrow);
+  a.n[3] = subb(a.n[3], KV_P1 & mask, borrow);
+}
+
+__device__ __forceinline__ void fe_add(fe &r, const fe &a, const fe &b) {
+  u64 carry = 0;
+  r.n[0] = addc(a.n[0], b.n[0], carry);
+  r.n[1] = addc(a.n[1], b.n[1], carry);
+  r.n[2] = addc(a.n[2], b.n[2], carry);
+  r.n[3] = addc(a.n[3], b.n[3], carry);
+  /* wrap: += carry * PC (carry 0/1) */
+  u64 add0 = carry * KV_PC;
+  u64 c2 = 0;
+  r.n[0] = addc(r.n[0], add0, c2);
+  r.n[1] = addc(r.n[1], 0, c2);
+  r.n[2] = addc(r.n[2], 0, c2);
+  r.n[3] = addc(r.n[3], 0, c2);
+  /* c2 can only be set if r wrapped again with tiny value; fold once more */
+  u64 add1 = c2 * KV_PC;
+  u64 c3 = 0;
+  r.n[0] = addc(r.n[0], add1, c3);
+  r.n[1] = addc(r.n[1], 0, c3);
+  r.n[2] = addc(r.n[2], 0, c3);
+  r.n[3] = addc(r.n[3], 0, c3);
+  fe_norm_once(r);
+}
+
+__device__ __forceinline__ void fe_neg(fe &r, const fe &a) {
+  /* p - a for a < p; a==0 → 0 */
+  u64 is_zero = ((a.n[0] | a.n[1] | a.n[2] | a.n[3]) == 0);
+  u64 borrow = 0;
+  r.n[0] = subb(KV_P0, a.n[0], borrow);
+  r.n[1] = subb(KV_P1, a.n[1], borrow);
+  r.n[2] = subb(KV_P1, a.n[2], borrow);
+  r.n[3] = subb(KV_P1, a.n[3], borrow);
+  u64 mask = 0 - is_zero; /* if a==0 result must be 0 not p */
+  r.n[0] &= ~mask;
+  r.n[1] &= ~mask;
+  r.n[2] &= ~mask;
+  r.n[3] &= ~mask;
+}
+
+__device__ __forceinline__ void fe_sub(fe &r, const fe &a, const fe &b) {
+  fe nb;
+  fe_neg(nb, b);
+  fe_add(r, a, nb);
+}
+
+/* full 4x4 multiply (row-wise carry chain) + two-fold reduction via
+ * 2^256 ≡ PC (mod p). Straight-line, branch-free. */
+__device__ __forceinline__ void fe_mul_inner(u64 t[8], const u64 *an, const u64 *bn) {
+#pragma unroll
+  for (int i = 0; i < 8; i++) t[i] = 0;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    u64 carry = 0;
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      u64 lo = an[i] * bn[j];
+      u64 hi = __umul64hi(an[i], bn[j]);
+      u64 c = 0;
+      t[i + j] = addc(t[i + j], lo, c);
+      u64 c2 = 0;
+      t[i + j] = addc(t[i + j], carry, c2);
+      carry = hi + c + c2; /* hi <= 2^64-2: no overflow */
+    }
+    t[i + 4] = carry;
+  }
+}
+
+__device__ __forceinline__ void fe_reduce8(fe &r, const u64 t[8]) {
+  /* fold 1: s[0..4] = t[0..3] + PC * t[4..7] (PC is 33 bits) */
+  u64 lo0 = t[4] * KV_PC, hi0 = __umul64hi(t[4], KV_PC);
+  u64 lo1 = t[5] * KV_PC, hi1 = __umul64hi(t[5], KV_PC);
+  u64 lo2 = t[6] * KV_PC, hi2 = __umul64hi(t[6], KV_PC);
+  u64 lo3 = t[7] * KV_PC, hi3 = __umul64hi(t[7], KV_PC);
+  u64 cA = 0, cB = 0;
+  u64 s0 = addc(t[0], lo0, cA);
+  u64 s1 = addc(t[1], lo1, cA);
+  u64 s2 = addc(t[2], lo2, cA);
+  u64 s3 = addc(t[3], lo3, cA);
+  u64 s4 = cA;
+  s1 = addc(s1, hi0, cB);
+  s2 = addc(s2, hi1, cB);
+  s3 = addc(s3, hi2, cB);
+  s4 = s4 + hi3 + cB; /* < 2^34 */
+  /* fold 2: r = s[0..3] + PC * s4 */
+  u64 lo = s4 * KV_PC, hi = __umul64hi(s4, KV_PC);
+  u64 c = 0;
+  r.n[0] = addc(s0, lo, c);
+  r.n[1] = addc(s1, hi, c);
+  r.n[2] = addc(s2, 0, c);
+  r.n[3] = addc(s3, 0, c);
+  /* c set → value wrapped 2^256 exactly once and is now tiny (< 2^68) */
+  u64 add1 = c * KV_PC;
+  u64 c2 = 0;
+  r.n[0] = addc(r.n[0], add1, c2);
+  r.n[1] = addc(r.n[1], 0, c2);
+  r.n[2] = addc(r.n[2], 0, c2);
+  r.n[3] = addc(r.n[3], 0, c2);
+  fe_norm_once(r);
+}
+
+__device__ __forceinline__ void fe_mul(fe &r, const fe &a, const fe &b) {
+  u64 t[8];
+  fe_mul_inner(t, a.n, b.n);
+  fe_reduce8(r, t);
+}
+
+__device__ __forceinline__ void fe_sqr(fe &r, const fe &a) { fe_mul(r, a, a); }
+
+__device__ __forceinline__ void fe_mul_small(fe &r, const fe &a, u64 k) {
+  u64 c = 0, t4;
+  u64 lo0 = a.n[0] * k, hi0 = __umul64hi(a.n[0], k);
+  u64 lo1 = a.n[1] * k, hi1 = __umul64hi(a.n[1], k);
+  u64 lo2 = a.n[2] * k, hi2 = __umul64hi(a.n[2], k);
+  u64 lo3 = a.n[3] * k, hi3 = __umul64hi(a.n[3], k);
+  fe s;
+  s.n[0] = lo0;
+  s.n[1] = addc(lo1, hi0, c);
+  s.n[2] = addc(lo2, hi1, c);
+  s.n[3] = addc(lo3, hi2, c);
+  t4 = hi3 + c;
+  u64 lo = t4 * KV_PC, hi = __umul64hi(t4, KV_PC);
+  c = 0;
+  s.n[0] = addc(s.n[0], lo, c);
+  s.n[1] = addc(s.n[1], hi, c);
+  s.n[2] = addc(s.n[2], 0, c);
+  s.n[3] = addc(s.n[3], 0, c);
+  u64 add1 = c * KV_PC;
+  u64 c2 = 0;
+  s.n[0] = addc(s.n[0], add1, c2);
+  s.n[1] = addc(s.n[1], 0, c2);
+  s.n[2] = addc(s.n[2], 0, c2);
+  s.n[3] = addc(s.n[3], 0, c2);
+  fe_norm_once(s);
+  r = s;
+}
+
+__device__ __forceinline__ int fe_is_zero(const fe &a) {
+  return (a.n[0] | a.n[1] | a.n[2] | a.n[3]) == 0;
+}
+
+__device__ __forceinline__ int fe_eq(const fe &a, const fe &b) {
+  return ((a.n[0] ^ b.n[0]) | (a.n[1] ^ b.n[1]) | (a.n[2] ^ b.n[2]) |
+          (a.n[3] ^ b.n[3])) == 0;
+}
+
+__device__ __forceinline__ void fe_cmov(fe &r, const fe &a, u64 cond) {
+  u64 mask = 0 - cond;
+#pragma unroll
+  for (int i = 0; i < 4; i++) r.n[i] = (r.n[i] & ~mask) | (a.n[i] & mask);
+}
+
+/* a^e for fixed 256-bit big-endian exponent (square-and-multiply, MSB first) */
+__device__ inline void fe_pow(fe &r, const fe &a, const u64 e[4] /* e[3]=MSW */) {
+  fe result = {{1, 0, 0, 0}};
+  fe base = a;
+  for (int w = 3; w >= 0; w--) {
+    for (int b = 63; b >= 0; b--) {
+      fe_sqr(result, result);
+      if ((e[w] >> b) & 1) fe_mul(result, result, base);
+    }
+  }
+  r = result;
+}
+
+__device__ inline void fe_inv(fe &r, const fe &a) {
+  static const u64 PM2[4] = {0xFFFFFFFEFFFFFC2DULL, 0xFFFFFFFFFFFFFFFFULL,
+                             0xFFFFFFFFFFFFFFFFULL, 0xFFFFFFFFFFFFFFFFULL};
+  fe_pow(r, a, PM2);
+}
+
+/* sqrt via a^((p+1)/4); returns 1 if r*r == a */
+__device__ inline int fe_sqrt(fe &r, const fe &a) {
+  static const u64 SQ[4] = {0xFFFFFFFFBFFFFF0CULL, 0xFFFFFFFFFFFFFFFFULL,
+                            0xFFFFFFFFFFFFFFFFULL, 0x3FFFFFFFFFFFFFFFULL};
+  fe cand, chk;
+  fe_pow(cand, a, SQ);
+  fe_sqr(chk, cand);
+  r = cand;
+  return fe_eq(chk, a);
+}
+
+/* ---------- scalar mod n ---------- */
+
+struct sc {
+  u64 d[4];
+};
+
+#define KV_N0 0xBFD25E8CD0364141ULL
+#define KV_N1 0xBAAEDCE6AF48A03BULL
+#define KV_N2 0xFFFFFFFFFFFFFFFEULL
+#define KV_N3 0xFFFFFFFFFFFFFFFFULL
+/* 2^256 - n (129 bits, limbs) */
+#define KV_NC0 0x402DA1732FC9BEBFULL
+#define KV_NC1 0x4551231950B75FC4ULL
+#define KV_NC2 1ULL
+
+__device__ __forceinline__ int sc_gte_n(const sc &a) {
+  if (a.d[3] > KV_N3) return 1;
+  if (a.d[3] < KV_N3) return 0;
+  if (a.d[2] > KV_N2) return 1;
+  if (a.d[2] < KV_N2) return 0;
+  if (a.d[1] > KV_N1) return 1;
+  if (a.d[1] < KV_N1) return 0;
+  return a.d[0] >= KV_N0;
+}
+
+__device__ __forceinline__ void sc_sub_n(sc &a) {
+  u64 borrow = 0;
+  a.d[0] = subb(a.d[0], KV_N0, borrow);
+  a.d[1] = subb(a.d[1], KV_N1, borrow);
+  a.d[2] = subb(a.d[2], KV_N2, borrow);
+  a.d[3] = subb(a.d[3], KV_N3, borrow);
+}
+
+__device__ __forceinline__ int sc_is_zero(const sc &a) {
+  return (a.d[0] | a.d[1] | a.d[2] | a.d[3]) == 0;
+}
+
+/* from big-endian bytes; returns 1 on overflow (input >= n) */
+__device__ __forceinline__ int sc_from_be(sc &r, const uint8_t b[32]) {
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    u64 w = 0;
+#pragma unroll
+    for (int j = 0; j < 8; j++) w = (w << 8) | b[8 * (3 - i) + j];
+    r.d[i] = w;
+  }
+  int ov = sc_gte_n(r);
+  if (ov) sc_sub_n(r);
+  return ov;
+}
+
+__device__ __forceinline__ void sc_neg(sc &r, const sc &a) {
+  u64 is_zero = sc_is_zero(a);
+  u64 borrow = 0;
+  r.d[0] = subb(KV_N0, a.d[0], borrow);
+  r.d[1] = subb(KV_N1, a.d[1], borrow);
+  r.d[2] = subb(KV_N2, a.d[2], borrow);
+  r.d[3] = subb(KV_N3, a.d[3], borrow);
+  u64 mask = 0 - is_zero;
+  r.d[0] &= ~mask;
+  r.d[1] &= ~mask;
+  r.d[2] &= ~mask;
+  r.d[3] &= ~mask;
+}
+
+/* 512-bit -> mod n via repeated fold with NC (3 limbs) */
+__device__ inline void sc_reduce8(sc &r, const u64 tin[8]) {
+  u64 v[9];
+#pragma unroll
+  for (int i = 0; i < 8; i++) v[i] = tin[i];
+  v[8] = 0;
+  const u64 NC[3] = {KV_NC0, KV_NC1, KV_NC2};
+  /* three folds suffice: 8→7→6→5→(≈4+carry) then subtracts */
+  for (int pass = 0; pass < 4; pass++) {
+    u64 acc[9] = {v[0], v[1], v[2], v[3], 0, 0, 0, 0, 0};
+    int any_hi = 0;
+    for (int i = 4; i < 9; i++) any_hi |= (v[i] != 0);
+    if (!any_hi) break;
+    for (int i = 0; i < 5; i++) {
+      u64 hi_limb = v[4 + i];
+      if (!hi_limb) continue;
+      u64 carry = 0;
+      for (int j = 0; j < 3; j++) {
+        u64 lo = hi_limb * NC[j];
+        u64 h = __umul64hi(hi_limb, NC[j]);
+        u64 c = 0;
+        acc[i + j] = addc(acc[i + j], lo, c);
+        u64 c2 = 0;
+        acc[i + j + 1] = addc(acc[i + j + 1], h + carry, c2);
+        carry = c + c2;
+      }
+      int k = i + 4;
+      while (carry && k < 9) {
+        u64 c = 0;
+        acc[k] = addc(acc[k], carry, c);
+        carry = c;
+        k++;
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < 9; i++) v[i] = acc[i];
+  }
+  r.d[0] = v[0];
+  r.d[1] = v[1];
+  r.d[2] = v[2];
+  r.d[3] = v[3];
+  while (sc_gte_n(r)) sc_sub_n(r);
+}
+
+__device__ inline void sc_mul(sc &r, const sc &a, const sc &b) {
+  u64 t[8];
+  fe_mul_inner(t, a.d, b.d);
+  sc_reduce8(r, t);
+}
+
+/* scalar inverse mod n via Fermat (ECDSA only) */
+__device__ inline void sc_inv(sc &r, const sc &a) {
+  static const u64 NM2[4] = {0xBFD25E8CD036413FULL, 0xBAAEDCE6AF48A03BULL,
+                             0xFFFFFFFFFFFFFFFEULL, 0xFFFFFFFFFFFFFFFFULL};
+  sc result = {{1, 0, 0, 0}};
+  sc base = a;
+  for (int w = 3; w >= 0; w--) {
+    for (int b = 63; b >= 0; b--) {
+      sc_mul(result, result, result);
+      if ((NM2[w] >> b) & 1) sc_mul(result, result, base);
+    }
+  }
+  r = result;
+}
+
+/* ---------- group: Jacobian points, a=0 b=7 curve ---------- */
+
+struct ge {
+  fe x, y; /* affine */
+};
+
+struct gej {
+  fe x, y, z;
+  int infinity;
+};
+
+__device__ __forceinline__ void gej_set_infinity(gej &r) {
+  r.x = {{0, 0, 0, 0}};
+  r.y = {{1, 0, 0, 0}};
+  r.z = {{0, 0, 0, 0}};
+  r.infinity = 1;
+}
+
+__device__ inline void gej_double(gej &r, const gej &a) {
+  if (a.infinity) {
+    r = a;
+    return;
+  }
+  fe A, B, C, D, E, F, t;
+  fe_sqr(A, a.x);
+  fe_sqr(B, a.y);
+  fe_sqr(C, B);
+  fe_add(t, a.x, B);
+  fe_sqr(t, t);
+  fe_sub(t, t, A);
+  fe_sub(t, t, C);
+  fe_add(D, t, t);
+  fe_mul_small(E, A, 3);
+  fe_sqr(F, E);
+  fe_sub(r.x, F, D);
+  fe_sub(r.x, r.x, D);
+  fe_sub(t, D, r.x);
+  fe_mul(t, E, t);
+  fe C8;
+  fe_mul_small(C8, C, 8);
+  fe ny;
+  fe_sub(ny, t, C8);
+  fe_mul(t, a.y, a.z);
+  fe_add(r.z, t, t);
+  r.y = ny;
+  r.infinity = 0;
+}
+
+/* mixed add: r = a + B(affine). Handles a=inf, equal-x cases. */
+__device__ inline void gej_add_ge(gej &r, const gej &a, const ge &b) {
+  if (a.infinity) {
+    r.x = b.x;
+    r.y = b.y;
+    r.z = {{1, 0, 0, 0}};
+    r.infinity = 0;
+    return;
+  }
+  fe z1z1, u2, s2, h, hh, i, j, rr, v, t;
+  fe_sqr(z1z1, a.z);
+  fe_mul(u2, b.x, z1z1);
+  fe_mul(s2, b.y, a.z);
+  fe_mul(s2, s2, z1z1);
+  fe_sub(h, u2, a.x);  /* H = U2 - X1 */
+  fe_sub(rr, s2, a.y); /* r' = S2 - Y1 */
+  if (fe_is_zero(h)) {
+    if (fe_is_zero(rr)) {
+      gej_double(r, a);
+      return;
+    }
+    gej_set_infinity(r);
+    return;
+  }
+  fe_sqr(hh, h);
+  fe_add(i, hh, hh);
+  fe_add(i, i, i); /* I = 4H^2 */
+  fe_mul(j, h, i); /* J = H*I */
+  fe_add(rr, rr, rr); /* r = 2(S2-Y1) */
+  fe_mul(v, a.x, i);  /* V = X1*I */
+  fe_sqr(r.x, rr);
+  fe_sub(r.x, r.x, j);
+  fe_sub(r.x, r.x, v);
+  fe_sub(r.x, r.x, v);
+  fe_sub(t, v, r.x);
+  fe_mul(t, rr, t);
+  fe y1j;
+  fe_mul(y1j, a.y, j);
+  fe_add(y1j, y1j, y1j);
+  fe ny;
+  fe_sub(ny, t, y1j);
+  fe zz;
+  fe_add(zz, a.z, h);
+  fe_sqr(zz, zz);
+  fe_sub(zz, zz, z1z1);
+  fe_sub(zz, zz, hh);
+  r.z = zz;
+  r.y = ny;
+  r.infinity = 0;
+}
+
+} // namespace kv
+
+#endif /* KV_SECP_DEVICE_H */

@@ -1,0 +1,249 @@
+/* MI355X-native batched signature verification kernels (gfx950, PRODUCT code).
+ *
+ * kv_schnorr_verify_kernel ⇔ secp256k1::schnorr::Signature::verify at
+ *   crypto/txscript/src/lib.rs:869 (BIP-340), one signature per lane.
+ * kv_ecdsa_verify_kernel   ⇔ secp256k1::ecdsa::Signature::verify at
+ *   crypto/txscript/src/lib.rs:899 (low-S enforced like libsecp256k1).
+ *
+ * Layout: tuples are 128B (Schnorr: r‖s‖pk_x‖msg) / 129B-padded-to-132
+ * (ECDSA: r‖s‖pk33‖msg, see engine) records in HBM; loads are a handful of
+ * dwordx4 per lane against ~5k 256-bit field multiplies of ALU work — this
+ * kernel is integer-ALU bound by ~300 ops/byte, so layout
+ * matters far less than VGPR pressure and carry-chain depth.
+ *
+ * Verdicts are wave-ballot-compressed: each 64-lane wave writes one u64 of the
+ * bitmap with a single store (no atomics).
+ */
+#include "kv_hash_device.h"
+#include "kv_secp_device.h"
+
+namespace kv {
+
+/* SHA-256 midstate after compressing tag_hash‖tag_hash for
+ * tag = "BIP0340/challenge" (fixed by BIP-340; precomputed, verified against a
+ * full SHA-256 at build time in tests). */
+__device__ __constant__ static const uint32_t BIP340_CHALLENGE_MID[8] = {
+    0x9cecba11u, 0x23925381u, 0x11679112u, 0xd1627e0fu,
+    0x97c87550u, 0x003cc765u, 0x90f61164u, 0x33e9b66au};
+
+/* status codes (engine-internal, align with KV_SCRIPT_* mapping in host) */
+enum : uint8_t {
+  KVS_VALID = 0,
+  KVS_INVALID = 1,
+  KVS_BAD_PUBKEY = 2,
+  KVS_BAD_SIG = 3,
+};
+
+__device__ __forceinline__ void fe_from_be(fe &r, const uint8_t b[32]) {
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    u64 w = 0;
+#pragma unroll
+    for (int j = 0; j < 8; j++) w = (w << 8) | b[8 * (3 - i) + j];
+    r.n[i] = w;
+  }
+}
+
+__device__ __forceinline__ int fe_gte_p_full(const fe &a) {
+  return (a.n[3] == KV_P1) & (a.n[2] == KV_P1) & (a.n[1] == KV_P1) &
+         (a.n[0] >= KV_P0);
+}
+
+/* lift x (BE bytes) to even-y affine point; 0 on failure */
+__device__ inline int lift_x_even(ge &P, const uint8_t xb[32]) {
+  fe x;
+  fe_from_be(x, xb);
+  if (fe_gte_p_full(x)) return 0;
+  fe x3, y2, y;
+  fe_sqr(x3, x);
+  fe_mul(x3, x3, x);
+  fe seven = {{7, 0, 0, 0}};
+  fe_add(y2, x3, seven);
+  if (!fe_sqrt(y, y2)) return 0;
+  /* even y */
+  fe ny;
+  fe_neg(ny, y);
+  fe_cmov(y, ny, y.n[0] & 1);
+  P.x = x;
+  P.y = y;
+  return 1;
+}
+
+/* parse 33-byte compressed pubkey; 0 on failure */
+__device__ inline int parse_compressed(ge &P, const uint8_t pk[33]) {
+  if (pk[0] != 0x02 && pk[0] != 0x03) return 0;
+  fe x;
+  fe_from_be(x, pk + 1);
+  if (fe_gte_p_full(x)) return 0;
+  fe x3, y2, y;
+  fe_sqr(x3, x);
+  fe_mul(x3, x3, x);
+  fe seven = {{7, 0, 0, 0}};
+  fe_add(y2, x3, seven);
+  if (!fe_sqrt(y, y2)) return 0;
+  fe ny;
+  fe_neg(ny, y);
+  u64 want_odd = (pk[0] == 0x03);
+  fe_cmov(y, ny, (y.n[0] & 1) ^ want_odd);
+  P.x = x;
+  P.y = y;
+  return 1;
+}
+
+/* affine G */
+__device__ __constant__ static const ge GE_G = {
+    {{0x59F2815B16F81798ULL, 0x029BFCDB2DCE28D9ULL, 0x55A06295CE870B07ULL,
+      0x79BE667EF9DCBBACULL}},
+    {{0x9C47D08FFB10D4B8ULL, 0xFD17B448A6855419ULL, 0x5DA4FBFC0E1108A8ULL,
+      0x483ADA7726A3C465ULL}}};
+
+/* R = gs*G + ps*P — interleaved per-bit double-and-add (round-1 structure;
+ * windowed/LDS-table variants are the optimization target of later passes). */
+__device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const ge &P) {
+  gej_set_infinity(R);
+  int started = 0;
+  for (int w = 3; w >= 0; w--) {
+    u64 gw = gs.d[w], pw = ps.d[w];
+    for (int b = 63; b >= 0; b--) {
+      if (started) {
+        gej t;
+        gej_double(t, R);
+        R = t;
+      }
+      u64 gbit = (gw >> b) & 1, pbit = (pw >> b) & 1;
+      if (gbit) {
+        gej t;
+        gej_add_ge(t, R, GE_G);
+        R = t;
+        started = 1;
+      }
+      if (pbit) {
+        gej t;
+        gej_add_ge(t, R, P);
+        R = t;
+        started = 1;
+      }
+    }
+  }
+}
+
+/* One BIP-340 verification; returns KVS_* */
+__device__ inline uint8_t schnorr_verify_one(const uint8_t *rb, const uint8_t *sb,
+                                             const uint8_t *pkb, const uint8_t *msg) {
+  ge P;
+  if (!lift_x_even(P, pkb)) return KVS_BAD_PUBKEY;
+  fe rx;
+  fe_from_be(rx, rb);
+  if (fe_gte_p_full(rx)) return KVS_INVALID;
+  sc s;
+  if (sc_from_be(s, sb)) return KVS_INVALID;
+  uint8_t eh[32];
+  sha256_tagged96(BIP340_CHALLENGE_MID, rb, pkb, msg, eh);
+  sc e, ne;
+  sc_from_be(e, eh);
+  sc_neg(ne, e);
+  gej R;
+  ecmult_double(R, s, ne, P);
+  if (R.infinity) return KVS_INVALID;
+  /* affine via one inversion: need x == r and even y */
+  fe zi, zi2, zi3, xa, ya;
+  fe_inv(zi, R.z);
+  fe_sqr(zi2, zi);
+  fe_mul(zi3, zi2, zi);
+  fe_mul(xa, R.x, zi2);
+  fe_mul(ya, R.y, zi3);
+  if (ya.n[0] & 1) return KVS_INVALID;
+  return fe_eq(xa, rx) ? KVS_VALID : KVS_INVALID;
+}
+
+extern "C" __global__ void kv_schnorr_verify_kernel(const uint8_t *__restrict__ tuples,
+                                                    unsigned long long n,
+                                                    unsigned long long *__restrict__ bitmap,
+                                                    uint8_t *__restrict__ status) {
+  unsigned long long i = (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  int valid = 0;
+  if (i < n) {
+    const uint8_t *t = tuples + i * 128;
+    uint8_t st = schnorr_verify_one(t, t + 32, t + 64, t + 96);
+    if (status) status[i] = st;
+    valid = (st == KVS_VALID);
+  }
+  unsigned long long mask = __ballot(valid);
+  if ((threadIdx.x & 63) == 0) {
+    unsigned long long word = i / 64; /* lane0's index is 64-aligned */
+    if (i < n) bitmap[word] = mask;
+  }
+}
+
+/* ---------------- ECDSA ---------------- */
+
+__device__ inline uint8_t ecdsa_verify_one(const uint8_t *rb, const uint8_t *sb,
+                                           const uint8_t *pk33, const uint8_t *msg) {
+  ge P;
+  if (!parse_compressed(P, pk33)) return KVS_BAD_PUBKEY;
+  sc r, s;
+  if (sc_from_be(r, rb)) return KVS_BAD_SIG;  /* parse_compact overflow */
+  if (sc_from_be(s, sb)) return KVS_BAD_SIG;
+  if (sc_is_zero(r) || sc_is_zero(s)) return KVS_INVALID;
+  /* libsecp256k1 secp256k1_ecdsa_verify rejects high-S */
+  static const u64 NHALF[4] = {0xDFE92F46681B20A0ULL, 0x5D576E7357A4501DULL,
+                               0xFFFFFFFFFFFFFFFFULL, 0x7FFFFFFFFFFFFFFFULL};
+  int high = 0;
+  for (int i = 3; i >= 0; i--) {
+    if (s.d[i] > NHALF[i]) { high = 1; break; }
+    if (s.d[i] < NHALF[i]) break;
+  }
+  if (high) return KVS_INVALID;
+  sc z;
+  sc_from_be(z, msg); /* reduced mod n like secp256k1_scalar_set_b32 */
+  sc w, u1, u2;
+  sc_inv(w, s);
+  sc_mul(u1, z, w);
+  sc_mul(u2, r, w);
+  gej R;
+  ecmult_double(R, u1, u2, P);
+  if (R.infinity) return KVS_INVALID;
+  /* x(R) ≡ r (mod n): X == (r + k·n)·Z² for k ∈ {0,1} with r+n < p */
+  fe z2;
+  fe_sqr(z2, R.z);
+  uint8_t rbe[32];
+#pragma unroll
+  for (int i = 0; i < 4; i++)
+#pragma unroll
+    for (int j = 0; j < 8; j++) rbe[8 * (3 - i) + j] = (uint8_t)(r.d[i] >> (56 - 8 * j));
+  fe rf;
+  fe_from_be(rf, rbe);
+  fe t;
+  fe_mul(t, rf, z2);
+  if (fe_eq(t, R.x)) return KVS_VALID;
+  /* r + n */
+  u64 carry = 0;
+  fe rn;
+  rn.n[0] = addc(rf.n[0], KV_N0, carry);
+  rn.n[1] = addc(rf.n[1], KV_N1, carry);
+  rn.n[2] = addc(rf.n[2], KV_N2, carry);
+  rn.n[3] = addc(rf.n[3], KV_N3, carry);
+  if (!carry && !fe_gte_p_full(rn)) {
+    fe_mul(t, rn, z2);
+    if (fe_eq(t, R.x)) return KVS_VALID;
+  }
+  return KVS_INVALID;
+}
+
+extern "C" __global__ void kv_ecdsa_verify_kernel(const uint8_t *__restrict__ tuples,
+                                                  unsigned long long n,
+                                                  unsigned long long *__restrict__ bitmap,
+                                                  uint8_t *__restrict__ status) {
+  unsigned long long i = (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+  int valid = 0;
+  if (i < n) {
+    const uint8_t *t = tuples + i * 132; /* r32‖s32‖pk33‖msg32‖pad3 */
+    uint8_t st = ecdsa_verify_one(t, t + 32, t + 64, t + 97);
+    if (status) status[i] = st;
+    valid = (st == KVS_VALID);
+  }
+  unsigned long long mask = __ballot(valid);
+  if ((threadIdx.x & 63) == 0 && i < n) bitmap[i / 64] = mask;
+}
+
+} // namespace kv
