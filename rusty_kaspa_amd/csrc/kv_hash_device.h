@@ -8,7 +8,9 @@
 #ifndef KV_HASH_DEVICE_H
 #define KV_HASH_DEVICE_H
 
+#ifndef KV_HOST_TEST
 #include <hip/hip_runtime.h>
+#endif
 #include <stdint.h>
 
 namespace kv {

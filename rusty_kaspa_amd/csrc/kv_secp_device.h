@@ -16,7 +16,9 @@
 #ifndef KV_SECP_DEVICE_H
 #define KV_SECP_DEVICE_H
 
+#ifndef KV_HOST_TEST
 #include <hip/hip_runtime.h>
+#endif
 #include <stdint.h>
 
 namespace kv {
@@ -324,47 +326,74 @@ __device__ __forceinline__ void sc_neg(sc &r, const sc &a) {
   r.d[3] &= ~mask;
 }
 
-/* 512-bit -> mod n via repeated fold with NC (3 limbs) */
+/* 512-bit → mod n: three bounded folds with NC = 2^256 - n (129 bits, 3 limbs).
+ * Each fold: value = lo(4 limbs) + NC × hi(k limbs); widths shrink 4→3→1→ε. */
 __device__ inline void sc_reduce8(sc &r, const u64 tin[8]) {
-  u64 v[9];
-#pragma unroll
-  for (int i = 0; i < 8; i++) v[i] = tin[i];
-  v[8] = 0;
   const u64 NC[3] = {KV_NC0, KV_NC1, KV_NC2};
-  /* three folds suffice: 8→7→6→5→(≈4+carry) then subtracts */
-  for (int pass = 0; pass < 4; pass++) {
-    u64 acc[9] = {v[0], v[1], v[2], v[3], 0, 0, 0, 0, 0};
-    int any_hi = 0;
-    for (int i = 4; i < 9; i++) any_hi |= (v[i] != 0);
-    if (!any_hi) break;
-    for (int i = 0; i < 5; i++) {
-      u64 hi_limb = v[4 + i];
-      if (!hi_limb) continue;
-      u64 carry = 0;
-      for (int j = 0; j < 3; j++) {
-        u64 lo = hi_limb * NC[j];
-        u64 h = __umul64hi(hi_limb, NC[j]);
-        u64 c = 0;
-        acc[i + j] = addc(acc[i + j], lo, c);
-        u64 c2 = 0;
-        acc[i + j + 1] = addc(acc[i + j + 1], h + carry, c2);
-        carry = c + c2;
-      }
-      int k = i + 4;
-      while (carry && k < 9) {
-        u64 c = 0;
-        acc[k] = addc(acc[k], carry, c);
-        carry = c;
-        k++;
-      }
+  /* fold 1: s[7] = t[0..3] + NC × t[4..7] */
+  u64 s[7] = {tin[0], tin[1], tin[2], tin[3], 0, 0, 0};
+  for (int i = 0; i < 4; i++) {
+    u64 carry = 0;
+    for (int j = 0; j < 3; j++) {
+      u64 lo = tin[4 + i] * NC[j];
+      u64 hi = __umul64hi(tin[4 + i], NC[j]);
+      u64 c = 0;
+      s[i + j] = addc(s[i + j], lo, c);
+      u64 c2 = 0;
+      s[i + j] = addc(s[i + j], carry, c2);
+      carry = hi + c + c2;
     }
-#pragma unroll
-    for (int i = 0; i < 9; i++) v[i] = acc[i];
+    /* deposit final carry at s[i+3] and ripple */
+    u64 c = 0;
+    s[i + 3] = addc(s[i + 3], carry, c);
+    for (int k = i + 4; k < 7 && c; k++) s[k] = addc(s[k], 0, c);
   }
-  r.d[0] = v[0];
-  r.d[1] = v[1];
-  r.d[2] = v[2];
-  r.d[3] = v[3];
+  /* fold 2: u[5] = s[0..3] + NC × s[4..6] */
+  u64 u[5] = {s[0], s[1], s[2], s[3], 0};
+  for (int i = 0; i < 3; i++) {
+    u64 carry = 0;
+    for (int j = 0; j < 3; j++) {
+      int pos = i + j;
+      u64 lo = s[4 + i] * NC[j];
+      u64 hi = __umul64hi(s[4 + i], NC[j]);
+      u64 c = 0;
+      u[pos] = addc(u[pos], lo, c);
+      u64 c2 = 0;
+      u[pos] = addc(u[pos], carry, c2);
+      carry = hi + c + c2;
+    }
+    /* deposit final carry; for i==2 the carry is provably 0 (value < 2^259) */
+    if (i + 3 < 5) {
+      u64 c = 0;
+      u[i + 3] = addc(u[i + 3], carry, c);
+      for (int k = i + 4; k < 5 && c; k++) u[k] = addc(u[k], 0, c);
+    }
+  }
+  /* fold 3: w[4] + carry = u[0..3] + NC × u[4]  (u[4] small) */
+  u64 w[4] = {u[0], u[1], u[2], u[3]};
+  {
+    u64 carry = 0;
+    for (int j = 0; j < 3; j++) {
+      u64 lo = u[4] * NC[j];
+      u64 hi = __umul64hi(u[4], NC[j]);
+      u64 c = 0;
+      w[j] = addc(w[j], lo, c);
+      u64 c2 = 0;
+      w[j] = addc(w[j], carry, c2);
+      carry = hi + c + c2;
+    }
+    u64 c = 0;
+    w[3] = addc(w[3], carry, c);
+    /* fold 4: carry bit ⇒ += NC once more (cannot carry again: value < 2^130+ε) */
+    if (c) {
+      u64 cc = 0;
+      w[0] = addc(w[0], NC[0], cc);
+      w[1] = addc(w[1], NC[1], cc);
+      w[2] = addc(w[2], NC[2], cc);
+      w[3] = addc(w[3], 0, cc);
+    }
+  }
+  r.d[0] = w[0]; r.d[1] = w[1]; r.d[2] = w[2]; r.d[3] = w[3];
   while (sc_gte_n(r)) sc_sub_n(r);
 }
 
