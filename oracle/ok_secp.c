@@ -16,6 +16,7 @@
  *   - schnorr verify: r >= p or s >= n → bool false (no parse error)
  */
 #include "oracle.h"
+#include <pthread.h>
 #include <string.h>
 
 typedef unsigned __int128 u128;
@@ -525,9 +526,15 @@ static int wnaf(int8_t out[257], const sc *a, int w) {
 #define G_TABLE_SIZE (1 << (G_WINDOW - 1)) /* 128 odd multiples: G,3G,...,255G */
 static ge g_table[G_TABLE_SIZE];
 static int g_table_ready = 0;
+static pthread_mutex_t g_table_mu = PTHREAD_MUTEX_INITIALIZER;
 
 static void ensure_g_table(void) {
   if (__atomic_load_n(&g_table_ready, __ATOMIC_ACQUIRE)) return;
+  pthread_mutex_lock(&g_table_mu);
+  if (__atomic_load_n(&g_table_ready, __ATOMIC_ACQUIRE)) {
+    pthread_mutex_unlock(&g_table_mu);
+    return;
+  }
   ge g = {GE_GX, GE_GY, 0};
   gej gj, g2;
   gej_set_ge(&gj, &g);
@@ -540,6 +547,7 @@ static void ensure_g_table(void) {
     cur = next;
   }
   __atomic_store_n(&g_table_ready, 1, __ATOMIC_RELEASE);
+  pthread_mutex_unlock(&g_table_mu);
 }
 
 /* r = gs*G + ps*P (either may be NULL) */

@@ -231,14 +231,21 @@ __device__ __forceinline__ void fe_cmov(fe &r, const fe &a, u64 cond) {
   for (int i = 0; i < 4; i++) r.n[i] = (r.n[i] & ~mask) | (a.n[i] & mask);
 }
 
-/* a^e for fixed 256-bit big-endian exponent (square-and-multiply, MSB first) */
-__device__ inline void fe_pow(fe &r, const fe &a, const u64 e[4] /* e[3]=MSW */) {
+/* a^e for fixed 256-bit big-endian exponent (square-and-multiply, MSB first).
+ * noinline + unroll-disable: a fully-unrolled 256-step square-and-multiply
+ * inflates the kernel by ~50k instructions and pushes it into long-branch
+ * relaxation territory. */
+__device__ __noinline__ void fe_pow(fe &r, const fe &a, const u64 e[4] /* e[3]=MSW */) {
   fe result = {{1, 0, 0, 0}};
   fe base = a;
+#pragma unroll 1
   for (int w = 3; w >= 0; w--) {
+#pragma unroll 1
     for (int b = 63; b >= 0; b--) {
       fe_sqr(result, result);
-      if ((e[w] >> b) & 1) fe_mul(result, result, base);
+      fe t;
+      fe_mul(t, result, base);
+      fe_cmov(result, t, (e[w] >> b) & 1);
     }
   }
   r = result;
@@ -404,43 +411,52 @@ __device__ inline void sc_mul(sc &r, const sc &a, const sc &b) {
 }
 
 /* scalar inverse mod n via Fermat (ECDSA only) */
-__device__ inline void sc_inv(sc &r, const sc &a) {
+__device__ __noinline__ void sc_inv(sc &r, const sc &a) {
   static const u64 NM2[4] = {0xBFD25E8CD036413FULL, 0xBAAEDCE6AF48A03BULL,
                              0xFFFFFFFFFFFFFFFEULL, 0xFFFFFFFFFFFFFFFFULL};
   sc result = {{1, 0, 0, 0}};
   sc base = a;
+#pragma unroll 1
   for (int w = 3; w >= 0; w--) {
+#pragma unroll 1
     for (int b = 63; b >= 0; b--) {
       sc_mul(result, result, result);
-      if ((NM2[w] >> b) & 1) sc_mul(result, result, base);
+      sc t;
+      sc_mul(t, result, base);
+      u64 bit = (NM2[w] >> b) & 1;
+      u64 mask = 0 - bit;
+      for (int k = 0; k < 4; k++) result.d[k] = (result.d[k] & ~mask) | (t.d[k] & mask);
     }
   }
   r = result;
 }
 
-/* ---------- group: Jacobian points, a=0 b=7 curve ---------- */
+/* ---------- group: Jacobian points, a=0 b=7 curve ----------
+ * Infinity is represented as Z == 0 (no flag field, no early returns): point
+ * structs stay fully SROA-decomposed into VGPRs — address-taken structs with
+ * control-flow copies were observed to fall back to flat memory on gfx950. */
 
 struct ge {
   fe x, y; /* affine */
 };
 
 struct gej {
-  fe x, y, z;
-  int infinity;
+  fe x, y, z; /* z == 0 ⇔ infinity */
 };
 
 __device__ __forceinline__ void gej_set_infinity(gej &r) {
-  r.x = {{0, 0, 0, 0}};
+  r.x = {{1, 0, 0, 0}};
   r.y = {{1, 0, 0, 0}};
   r.z = {{0, 0, 0, 0}};
-  r.infinity = 1;
 }
 
-__device__ inline void gej_double(gej &r, const gej &a) {
-  if (a.infinity) {
-    r = a;
-    return;
-  }
+__device__ __forceinline__ int gej_is_infinity(const gej &a) {
+  return fe_is_zero(a.z);
+}
+
+/* doubling: straight-line, valid for z==0 (result keeps z==0).
+ * y == 0 cannot occur on secp256k1 (no 2-torsion). */
+__device__ __noinline__ void gej_double(gej &r, const gej &a) {
   fe A, B, C, D, E, F, t;
   fe_sqr(A, a.x);
   fe_sqr(B, a.y);
@@ -452,69 +468,92 @@ __device__ inline void gej_double(gej &r, const gej &a) {
   fe_add(D, t, t);
   fe_mul_small(E, A, 3);
   fe_sqr(F, E);
-  fe_sub(r.x, F, D);
-  fe_sub(r.x, r.x, D);
-  fe_sub(t, D, r.x);
+  fe nx, ny, nz;
+  fe_sub(nx, F, D);
+  fe_sub(nx, nx, D);
+  fe_sub(t, D, nx);
   fe_mul(t, E, t);
   fe C8;
   fe_mul_small(C8, C, 8);
-  fe ny;
   fe_sub(ny, t, C8);
   fe_mul(t, a.y, a.z);
-  fe_add(r.z, t, t);
+  fe_add(nz, t, t);
+  r.x = nx;
   r.y = ny;
-  r.infinity = 0;
+  r.z = nz;
 }
 
-/* mixed add: r = a + B(affine). Handles a=inf, equal-x cases. */
-__device__ inline void gej_add_ge(gej &r, const gej &a, const ge &b) {
-  if (a.infinity) {
-    r.x = b.x;
-    r.y = b.y;
-    r.z = {{1, 0, 0, 0}};
-    r.infinity = 0;
-    return;
-  }
+/* mixed add r = a + B(affine): generic madd formulas computed unconditionally;
+ * a==infinity fixed up with selects; the rare equal-x cases (h==0 with a
+ * finite) take a divergent slow path only when a lane actually hits them. */
+__device__ __noinline__ void gej_add_ge(gej &r, const gej &a, const ge &b) {
+  u64 a_inf = (u64)fe_is_zero(a.z);
   fe z1z1, u2, s2, h, hh, i, j, rr, v, t;
   fe_sqr(z1z1, a.z);
   fe_mul(u2, b.x, z1z1);
   fe_mul(s2, b.y, a.z);
   fe_mul(s2, s2, z1z1);
-  fe_sub(h, u2, a.x);  /* H = U2 - X1 */
-  fe_sub(rr, s2, a.y); /* r' = S2 - Y1 */
-  if (fe_is_zero(h)) {
+  fe_sub(h, u2, a.x);
+  fe_sub(rr, s2, a.y);
+  if (!a_inf && fe_is_zero(h)) {
+    /* rare: same x. rr==0 → doubling; else opposite points → infinity */
     if (fe_is_zero(rr)) {
       gej_double(r, a);
-      return;
+    } else {
+      gej_set_infinity(r);
     }
-    gej_set_infinity(r);
     return;
   }
   fe_sqr(hh, h);
   fe_add(i, hh, hh);
-  fe_add(i, i, i); /* I = 4H^2 */
-  fe_mul(j, h, i); /* J = H*I */
-  fe_add(rr, rr, rr); /* r = 2(S2-Y1) */
-  fe_mul(v, a.x, i);  /* V = X1*I */
-  fe_sqr(r.x, rr);
-  fe_sub(r.x, r.x, j);
-  fe_sub(r.x, r.x, v);
-  fe_sub(r.x, r.x, v);
-  fe_sub(t, v, r.x);
+  fe_add(i, i, i);
+  fe_mul(j, h, i);
+  fe_add(rr, rr, rr);
+  fe_mul(v, a.x, i);
+  fe nx, ny, nz;
+  fe_sqr(nx, rr);
+  fe_sub(nx, nx, j);
+  fe_sub(nx, nx, v);
+  fe_sub(nx, nx, v);
+  fe_sub(t, v, nx);
   fe_mul(t, rr, t);
   fe y1j;
   fe_mul(y1j, a.y, j);
   fe_add(y1j, y1j, y1j);
-  fe ny;
   fe_sub(ny, t, y1j);
   fe zz;
   fe_add(zz, a.z, h);
   fe_sqr(zz, zz);
   fe_sub(zz, zz, z1z1);
   fe_sub(zz, zz, hh);
-  r.z = zz;
+  nz = zz;
+  /* a was infinity → result is b (z = 1) */
+  static const fe FE_ONE = {{1, 0, 0, 0}};
+  fe_cmov(nx, b.x, a_inf);
+  fe_cmov(ny, b.y, a_inf);
+  fe_cmov(nz, FE_ONE, a_inf);
+  r.x = nx;
   r.y = ny;
-  r.infinity = 0;
+  r.z = nz;
+}
+
+__device__ __forceinline__ void gej_cmov(gej &r, const gej &a, u64 cond) {
+  u64 mask = 0 - cond;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    r.x.n[i] = (r.x.n[i] & ~mask) | (a.x.n[i] & mask);
+    r.y.n[i] = (r.y.n[i] & ~mask) | (a.y.n[i] & mask);
+    r.z.n[i] = (r.z.n[i] & ~mask) | (a.z.n[i] & mask);
+  }
+}
+
+/* branch-light mixed add: computes the generic add formulas unconditionally and
+ * fixes up the special cases (a=inf, double, opposite) with selects. The rare
+ * truly-special cases (h==0) take a divergent slow path only when they occur. */
+__device__ inline void gej_add_ge_sel(gej &r, const gej &a, const ge &b) {
+  gej out;
+  gej_add_ge(out, a, b);
+  r = out;
 }
 
 } // namespace kv

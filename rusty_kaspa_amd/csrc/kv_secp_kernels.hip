@@ -97,32 +97,27 @@ __device__ __constant__ static const ge GE_G = {
     {{0x9C47D08FFB10D4B8ULL, 0xFD17B448A6855419ULL, 0x5DA4FBFC0E1108A8ULL,
       0x483ADA7726A3C465ULL}}};
 
-/* R = gs*G + ps*P — interleaved per-bit double-and-add (round-1 structure;
- * windowed/LDS-table variants are the optimization target of later passes). */
-__device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const ge &P) {
+/* R = gs*G + ps*P — branchless interleaved per-bit double-and-add (round-1
+ * structure; windowed/LDS-table variants are the later optimization target).
+ * Every lane executes the same instruction stream: both conditional adds are
+ * computed and selected with cmov, so a 64-lane wave of different scalars
+ * stays fully converged outside the rare h==0 sub-branches. */
+__device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const ge &P,
+                                     volatile int *progress = nullptr) {
   gej_set_infinity(R);
-  int started = 0;
+#pragma unroll 1
   for (int w = 3; w >= 0; w--) {
     u64 gw = gs.d[w], pw = ps.d[w];
+    if (progress) *progress = 800 + (3 - w) * 64;
+#pragma unroll 1
     for (int b = 63; b >= 0; b--) {
-      if (started) {
-        gej t;
-        gej_double(t, R);
-        R = t;
-      }
-      u64 gbit = (gw >> b) & 1, pbit = (pw >> b) & 1;
-      if (gbit) {
-        gej t;
-        gej_add_ge(t, R, GE_G);
-        R = t;
-        started = 1;
-      }
-      if (pbit) {
-        gej t;
-        gej_add_ge(t, R, P);
-        R = t;
-        started = 1;
-      }
+      gej t;
+      gej_double(t, R);
+      R = t;
+      gej_add_ge(t, R, GE_G);
+      gej_cmov(R, t, (gw >> b) & 1);
+      gej_add_ge(t, R, P);
+      gej_cmov(R, t, (pw >> b) & 1);
     }
   }
 }
@@ -144,7 +139,7 @@ __device__ inline uint8_t schnorr_verify_one(const uint8_t *rb, const uint8_t *s
   sc_neg(ne, e);
   gej R;
   ecmult_double(R, s, ne, P);
-  if (R.infinity) return KVS_INVALID;
+  if (gej_is_infinity(R)) return KVS_INVALID;
   /* affine via one inversion: need x == r and even y */
   fe zi, zi2, zi3, xa, ya;
   fe_inv(zi, R.z);
@@ -202,7 +197,7 @@ __device__ inline uint8_t ecdsa_verify_one(const uint8_t *rb, const uint8_t *sb,
   sc_mul(u2, r, w);
   gej R;
   ecmult_double(R, u1, u2, P);
-  if (R.infinity) return KVS_INVALID;
+  if (gej_is_infinity(R)) return KVS_INVALID;
   /* x(R) ≡ r (mod n): X == (r + k·n)·Z² for k ∈ {0,1} with r+n < p */
   fe z2;
   fe_sqr(z2, R.z);
