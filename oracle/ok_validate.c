@@ -112,8 +112,11 @@ static int validate_impl(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_
           continue;
         }
         uint64_t fee = 0;
-        int code = validate_one_tx(&tx, pov_daa_score, flags, DEFAULT_COINBASE_MATURITY,
-                                   DEFAULT_MASS_PER_SIG_OP, &fee);
+        int code = ok_tx_is_coinbase(&tx)
+                       ? KV_ERR_BAD_BLOB /* coinbase never enters this path */
+                       : validate_one_tx(&tx, pov_daa_score, flags,
+                                         DEFAULT_COINBASE_MATURITY,
+                                         DEFAULT_MASS_PER_SIG_OP, &fee);
         tx_codes_out[t] = code;
         fees_out[t] = fee;
         if (code == KV_OK && muhash_out)
@@ -142,8 +145,11 @@ static int validate_impl(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_
       continue;
     }
     uint64_t fee = 0;
-    int code = validate_one_tx(&tx, pov_daa_score, flags, DEFAULT_COINBASE_MATURITY,
-                               DEFAULT_MASS_PER_SIG_OP, &fee);
+    int code = ok_tx_is_coinbase(&tx)
+                   ? KV_ERR_BAD_BLOB /* coinbase never enters this path */
+                   : validate_one_tx(&tx, pov_daa_score, flags,
+                                     DEFAULT_COINBASE_MATURITY,
+                                     DEFAULT_MASS_PER_SIG_OP, &fee);
     tx_codes_out[t] = code;
     fees_out[t] = fee;
     if (code == KV_OK && muhash_out) ok_tx_muhash(&tx, block_daa_score, num, den);
