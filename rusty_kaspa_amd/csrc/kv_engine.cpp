@@ -1002,6 +1002,8 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
     }
   }
 
+  for (int t = 0; t < n_txs; t++)
+    if (codes[t]) fees[t] = 0; /* fee defined only for accepted txs */
   memcpy(tx_codes_out, codes.data(), (size_t)n_txs * 4);
   memcpy(fees_out, fees.data(), (size_t)n_txs * 8);
   return 0;
