@@ -121,9 +121,10 @@ def main():
         rc = lib.kv_verify_staged(ctx, ctypes.c_size_t(n), 0, ctypes.byref(kernel_ms))
         assert rc == 0, lib.kv_last_error().decode()
         if distributed:
-            # the real path's exchange: verdict bitmaps over RCCL (xGMI)
+            # the real path's exchange: every rank's verdict stripe is disjoint,
+            # so SUM == OR of the global verdict bitmap (RCCL over xGMI)
             import torch.distributed as dist
-            dist.all_reduce(bitmap_t, op=dist.ReduceOp.MIN)
+            dist.all_reduce(bitmap_t, op=dist.ReduceOp.SUM)
         return kernel_ms.value
 
     # warmup
