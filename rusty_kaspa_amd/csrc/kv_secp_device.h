@@ -23,6 +23,15 @@
 
 namespace kv {
 
+/* Build-time tuning knobs (perf experiments; default = the proven-safe config):
+ *  -DKV_GROUP_INLINE   : inline the group ops instead of __noinline__ calls
+ *                        (removes ABI spills; loops stay rolled via unroll 1) */
+#ifdef KV_GROUP_INLINE
+#define KV_GROUP_ATTR inline
+#else
+#define KV_GROUP_ATTR __noinline__
+#endif
+
 typedef uint64_t u64;
 typedef uint32_t u32;
 
@@ -235,7 +244,7 @@ __device__ __forceinline__ void fe_cmov(fe &r, const fe &a, u64 cond) {
  * noinline + unroll-disable: a fully-unrolled 256-step square-and-multiply
  * inflates the kernel by ~50k instructions and pushes it into long-branch
  * relaxation territory. */
-__device__ __noinline__ void fe_pow(fe &r, const fe &a, const u64 e[4] /* e[3]=MSW */) {
+__device__ KV_GROUP_ATTR void fe_pow(fe &r, const fe &a, const u64 e[4] /* e[3]=MSW */) {
   fe result = {{1, 0, 0, 0}};
   fe base = a;
 #pragma unroll 1
@@ -411,7 +420,7 @@ __device__ inline void sc_mul(sc &r, const sc &a, const sc &b) {
 }
 
 /* scalar inverse mod n via Fermat (ECDSA only) */
-__device__ __noinline__ void sc_inv(sc &r, const sc &a) {
+__device__ KV_GROUP_ATTR void sc_inv(sc &r, const sc &a) {
   static const u64 NM2[4] = {0xBFD25E8CD036413FULL, 0xBAAEDCE6AF48A03BULL,
                              0xFFFFFFFFFFFFFFFEULL, 0xFFFFFFFFFFFFFFFFULL};
   sc result = {{1, 0, 0, 0}};
@@ -456,7 +465,7 @@ __device__ __forceinline__ int gej_is_infinity(const gej &a) {
 
 /* doubling: straight-line, valid for z==0 (result keeps z==0).
  * y == 0 cannot occur on secp256k1 (no 2-torsion). */
-__device__ __noinline__ void gej_double(gej &r, const gej &a) {
+__device__ KV_GROUP_ATTR void gej_double(gej &r, const gej &a) {
   fe A, B, C, D, E, F, t;
   fe_sqr(A, a.x);
   fe_sqr(B, a.y);
@@ -486,7 +495,7 @@ __device__ __noinline__ void gej_double(gej &r, const gej &a) {
 /* mixed add r = a + B(affine): generic madd formulas computed unconditionally;
  * a==infinity fixed up with selects; the rare equal-x cases (h==0 with a
  * finite) take a divergent slow path only when a lane actually hits them. */
-__device__ __noinline__ void gej_add_ge(gej &r, const gej &a, const ge &b) {
+__device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
   u64 a_inf = (u64)fe_is_zero(a.z);
   fe z1z1, u2, s2, h, hh, i, j, rr, v, t;
   fe_sqr(z1z1, a.z);

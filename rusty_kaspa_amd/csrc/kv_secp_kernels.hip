@@ -151,7 +151,10 @@ __device__ inline uint8_t schnorr_verify_one(const uint8_t *rb, const uint8_t *s
   return fe_eq(xa, rx) ? KVS_VALID : KVS_INVALID;
 }
 
-extern "C" __global__ void kv_schnorr_verify_kernel(const uint8_t *__restrict__ tuples,
+#ifndef KV_LB
+#define KV_LB
+#endif
+extern "C" __global__ void KV_LB kv_schnorr_verify_kernel(const uint8_t *__restrict__ tuples,
                                                     unsigned long long n,
                                                     unsigned long long *__restrict__ bitmap,
                                                     uint8_t *__restrict__ status) {
@@ -225,7 +228,7 @@ __device__ inline uint8_t ecdsa_verify_one(const uint8_t *rb, const uint8_t *sb,
   return KVS_INVALID;
 }
 
-extern "C" __global__ void kv_ecdsa_verify_kernel(const uint8_t *__restrict__ tuples,
+extern "C" __global__ void KV_LB kv_ecdsa_verify_kernel(const uint8_t *__restrict__ tuples,
                                                   unsigned long long n,
                                                   unsigned long long *__restrict__ bitmap,
                                                   uint8_t *__restrict__ status) {
