@@ -1,31 +1,38 @@
-import ctypes, os, sys, time
+import ctypes, os, subprocess, sys
 os.chdir(os.path.dirname(os.path.abspath(__file__)))
-O = ctypes.CDLL("oracle/liboracle.so")
-n = 262144
-buf = ctypes.create_string_buffer(n*128)
-O.ok_gen_schnorr_tuples(ctypes.c_uint64(1), ctypes.c_size_t(n), 50, buf, 32)
-words = (n+63)//64
-exp = (ctypes.c_uint64 * words)()
-ncheck = 8192
-O.ok_verify_schnorr_batch(buf, ctypes.c_size_t(ncheck), 32, exp)
-for lib_name in sys.argv[1:]:
-    path = f"rusty_kaspa_amd/{lib_name}"
-    if not os.path.exists(path):
-        print(lib_name, "MISSING"); continue
-    L = ctypes.CDLL(path)
+
+if len(sys.argv) >= 3 and sys.argv[1] == "--one":
+    lib_name = sys.argv[2]
+    O = ctypes.CDLL("oracle/liboracle.so")
+    n = 262144
+    buf = ctypes.create_string_buffer(n*128)
+    O.ok_gen_schnorr_tuples(ctypes.c_uint64(1), ctypes.c_size_t(n), 50, buf, 32)
+    words = (n+63)//64
+    exp = (ctypes.c_uint64 * words)()
+    ncheck = 8192
+    O.ok_verify_schnorr_batch(buf, ctypes.c_size_t(ncheck), 32, exp)
+    L = ctypes.CDLL(f"rusty_kaspa_amd/{lib_name}")
     L.kv_create.restype = ctypes.c_void_p
     ctx = ctypes.c_void_p(L.kv_create(None))
-    assert ctx
     L.kv_stage_tuples(ctx, buf, ctypes.c_size_t(n), 0)
     ms = ctypes.c_double()
     times = []
     for _ in range(4):
-        rc = L.kv_verify_staged(ctx, ctypes.c_size_t(n), 0, ctypes.byref(ms))
-        assert rc == 0
+        assert L.kv_verify_staged(ctx, ctypes.c_size_t(n), 0, ctypes.byref(ms)) == 0
         times.append(ms.value)
     got = (ctypes.c_uint64 * words)()
     L.kv_fetch_bitmap(ctx, ctypes.c_size_t(n), got)
     ok = all(got[i] == exp[i] for i in range(ncheck//64))
     best = min(times[1:])
-    print(f"{lib_name}: {best:.2f} ms -> {n/best*1000/1e6:.2f} M/s parity={'OK' if ok else 'FAIL'}")
+    print(f"{lib_name}: {best:.2f} ms -> {n/best*1000/1e6:.2f} M/s parity={'OK' if ok else 'FAIL'}", flush=True)
     L.kv_destroy(ctx)
+    sys.exit(0)
+
+for lib in sys.argv[1:]:
+    try:
+        r = subprocess.run([sys.executable, "-u", __file__, "--one", lib],
+                           timeout=120, capture_output=True, text=True)
+        out = (r.stdout + r.stderr).strip()
+        print(out if out else f"{lib}: no output rc={r.returncode}", flush=True)
+    except subprocess.TimeoutExpired:
+        print(f"{lib}: TIMEOUT (hang)", flush=True)
