@@ -31,13 +31,13 @@ SEED = 42
 DEFAULT_TUPLES = 1 << 20  # 1M — the BASELINE config-2 batch
 INVALID_PERMILLE = 0  # all-valid variant is the headline; 10%-invalid via flag
 
-# Algorithmic work accounting for the roofline (documented in DESIGN.md §roofline):
-# per verify: 256 Jacobian doubles (7 fe_mul-equiv) + ~256 mixed adds (11) at
-# average density 0.5 each for G and P + x-lift sqrt (~380) + final inversion
-# (~380) ≈ 5,400 256-bit field multiplies; each fe_mul = 16 64×64 mults
-# (= 64 32×32 mult-equivalents) + fold ≈ 74 u32-mult-equiv + ~96 u32 add/carry
-# ≈ 170 u32 ALU ops → ≈ 0.92e6 u32-op-equivalents per verify.
-ALG_OPS_PER_VERIFY = 0.92e6
+# Algorithmic work accounting for the roofline (documented in DESIGN.md §4):
+# 4-bit windowed ladder: 256 Jacobian doubles (7 fe_mul-equiv) + 64 mixed adds
+# (11) + 64 full adds (16) + 15-entry P-table build (14×11) + x-lift sqrt
+# (~380) + final inversion (~380) ≈ 4,430 256-bit field multiplies; each
+# fe_mul ≈ 170 u32-ALU-op equivalents (64 32×32 mult-equiv + carries + fold)
+# → ≈ 0.75e6 u32-op-equivalents per verify.
+ALG_OPS_PER_VERIFY = 0.75e6
 # gfx950 VALU peak: 256 CU × 4 SIMD × 32 lanes × 2.4 GHz = 78.6 T u32-ops/s
 VALU_PEAK_TOPS = 78.6
 

@@ -87,6 +87,13 @@ extern "C" kv_ctx *kv_create(const kv_params *params) {
     delete ctx;
     return nullptr;
   }
+  /* fill the shared affine G multiples table (idempotent, once per device) */
+  hipLaunchKernelGGL(kv::kv_ec_table_init_kernel, dim3(1), dim3(64), 0, ctx->stream);
+  if (hipStreamSynchronize(ctx->stream) != hipSuccess) {
+    set_error("kv_create: G-table init failed");
+    delete ctx;
+    return nullptr;
+  }
   return ctx;
 }
 

@@ -546,6 +546,71 @@ __device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
   r.z = nz;
 }
 
+
+/* four doublings in one call — quarters the ABI crossings of the window ladder */
+__device__ KV_GROUP_ATTR void gej_double4(gej &r, const gej &a) {
+  gej t;
+  gej_double(t, a);
+  gej_double(r, t);
+  gej_double(t, r);
+  gej_double(r, t);
+}
+
+/* full Jacobian + Jacobian add (add-2007-bl shape), same select discipline as
+ * gej_add_ge. b with z==0 produces garbage that callers discard via cmov. */
+__device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
+  u64 a_inf = (u64)fe_is_zero(a.z);
+  fe z1z1, z2z2, u1, u2, s1, s2, h, i, j, rr, v, t;
+  fe_sqr(z1z1, a.z);
+  fe_sqr(z2z2, b.z);
+  fe_mul(u1, a.x, z2z2);
+  fe_mul(u2, b.x, z1z1);
+  fe_mul(s1, a.y, b.z);
+  fe_mul(s1, s1, z2z2);
+  fe_mul(s2, b.y, a.z);
+  fe_mul(s2, s2, z1z1);
+  fe_sub(h, u2, u1);
+  fe_sub(rr, s2, s1);
+  if (!a_inf && !fe_is_zero(b.z) && fe_is_zero(h)) {
+    if (fe_is_zero(rr)) {
+      gej_double(r, a);
+    } else {
+      gej_set_infinity(r);
+    }
+    return;
+  }
+  fe_add(rr, rr, rr); /* r = 2(S2-S1) */
+  fe_sqr(i, h);
+  fe_add(i, i, i);
+  fe_add(i, i, i); /* I = 4H^2 */
+  fe_mul(j, h, i);
+  fe_mul(v, u1, i);
+  fe nx, ny, nz;
+  fe_sqr(nx, rr);
+  fe_sub(nx, nx, j);
+  fe_sub(nx, nx, v);
+  fe_sub(nx, nx, v);
+  fe_sub(t, v, nx);
+  fe_mul(t, rr, t);
+  fe s1j;
+  fe_mul(s1j, s1, j);
+  fe_add(s1j, s1j, s1j);
+  fe_sub(ny, t, s1j);
+  fe zz;
+  fe_add(zz, a.z, b.z);
+  fe_sqr(zz, zz);
+  fe_sub(zz, zz, z1z1);
+  fe_sub(zz, zz, z2z2);
+  fe_mul(nz, zz, h);
+  /* a infinity → result = b */
+  fe_cmov(nx, b.x, a_inf);
+  fe_cmov(ny, b.y, a_inf);
+  fe_cmov(nz, b.z, a_inf);
+  r.x = nx;
+  r.y = ny;
+  r.z = nz;
+}
+
 __device__ __forceinline__ void gej_cmov(gej &r, const gej &a, u64 cond) {
   u64 mask = 0 - cond;
 #pragma unroll

@@ -97,28 +97,63 @@ __device__ __constant__ static const ge GE_G = {
     {{0x9C47D08FFB10D4B8ULL, 0xFD17B448A6855419ULL, 0x5DA4FBFC0E1108A8ULL,
       0x483ADA7726A3C465ULL}}};
 
-/* R = gs*G + ps*P — branchless interleaved per-bit double-and-add (round-1
- * structure; windowed/LDS-table variants are the later optimization target).
- * Every lane executes the same instruction stream: both conditional adds are
- * computed and selected with cmov, so a 64-lane wave of different scalars
- * stays fully converged outside the rare h==0 sub-branches. */
+/* Precomputed affine multiples 1..15 of G (filled once per device by
+ * kv_ec_table_init_kernel; entry 0 unused). Global memory: every lane reads
+ * the same 16 cached entries. */
+__device__ ge KV_G_TABLE[16];
+
+extern "C" __global__ void kv_ec_table_init_kernel() {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  gej acc;
+  acc.x = GE_G.x;
+  acc.y = GE_G.y;
+  acc.z = {{1, 0, 0, 0}};
+  for (int k = 1; k <= 15; k++) {
+    fe zi, zi2, zi3;
+    fe_inv(zi, acc.z);
+    fe_sqr(zi2, zi);
+    fe_mul(zi3, zi2, zi);
+    fe_mul(KV_G_TABLE[k].x, acc.x, zi2);
+    fe_mul(KV_G_TABLE[k].y, acc.y, zi3);
+    gej t;
+    gej_add_ge(t, acc, GE_G);
+    acc = t;
+  }
+}
+
+/* R = gs*G + ps*P — 4-bit fixed-window interleaved double-and-add.
+ * Per 4-bit window: 4 doublings + one mixed add against the shared affine
+ * G table + one full Jacobian add against the per-lane P table, each selected
+ * with cmov (digit 0 discards the add) so the wave stays converged.
+ * ≈ 256 doubles + 64 mixed + 64 full adds + 15-entry P-table build —
+ * ~1.8× fewer field multiplies than the per-bit form it replaces. */
 __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const ge &P,
                                      volatile int *progress = nullptr) {
+  /* P table (Jacobian): ptab[k] = k·P for k = 1..15 (mixed adds) */
+  gej ptab[16];
+  ptab[1].x = P.x;
+  ptab[1].y = P.y;
+  ptab[1].z = {{1, 0, 0, 0}};
+#pragma unroll 1
+  for (int k = 2; k <= 15; k++) gej_add_ge(ptab[k], ptab[k - 1], P);
+  gej_set_infinity(ptab[0]); /* never selected; keeps reads defined */
   gej_set_infinity(R);
+  if (progress) *progress = 800;
 #pragma unroll 1
-  for (int w = 3; w >= 0; w--) {
-    u64 gw = gs.d[w], pw = ps.d[w];
-    if (progress) *progress = 800 + (3 - w) * 64;
-#pragma unroll 1
-    for (int b = 63; b >= 0; b--) {
-      gej t;
-      gej_double(t, R);
-      R = t;
-      gej_add_ge(t, R, GE_G);
-      gej_cmov(R, t, (gw >> b) & 1);
-      gej_add_ge(t, R, P);
-      gej_cmov(R, t, (pw >> b) & 1);
-    }
+  for (int w = 63; w >= 0; w--) {
+    gej t;
+    gej_double(t, R);
+    gej_double(R, t);
+    gej_double(t, R);
+    gej_double(R, t);
+    int word = w >> 4, shift = (w & 15) * 4;
+    u64 dg = (gs.d[word] >> shift) & 15;
+    u64 dp = (ps.d[word] >> shift) & 15;
+    ge gent = KV_G_TABLE[dg];
+    gej_add_ge(t, R, gent);
+    gej_cmov(R, t, (u64)(dg != 0));
+    gej_add(t, R, ptab[dp]);
+    gej_cmov(R, t, (u64)(dp != 0));
   }
 }
 
@@ -152,7 +187,8 @@ __device__ inline uint8_t schnorr_verify_one(const uint8_t *rb, const uint8_t *s
 }
 
 #ifndef KV_LB
-#define KV_LB
+/* default: 2 waves/SIMD → 256-VGPR budget, zero spill in the ladder */
+#define KV_LB __launch_bounds__(256, 2)
 #endif
 extern "C" __global__ void KV_LB kv_schnorr_verify_kernel(const uint8_t *__restrict__ tuples,
                                                     unsigned long long n,
