@@ -80,6 +80,7 @@ extern "C" {
 #define KV_ERR_WRONG_MASS 5
 #define KV_ERR_SEQUENCE_LOCK 6
 #define KV_ERR_FEERATE_TOO_LOW 7
+#define KV_ERR_MASS_INCOMPUTABLE 8
 #define KV_ERR_BAD_BLOB 90
 #define KV_ERR_SIGNATURE_INVALID_BASE 100
 #define KV_ERR_SIGNATURE_EMPTY_BASE 200

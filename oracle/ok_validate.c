@@ -28,6 +28,80 @@
 #define DEFAULT_COINBASE_MATURITY 1000
 #define DEFAULT_MASS_PER_SIG_OP 1000
 
+
+/* ---------------- storage mass (KIP-0009), restating
+ * consensus/core/src/mass/mod.rs:385-514 + utxo_plurality (:86-105).
+ * storage_mass_parameter = STORAGE_MASS_PARAMETER (constants.rs:26) = 1e12. */
+#define STORM_PARAM (100000000ULL * 10000ULL)
+
+static uint64_t plurality(uint32_t spk_len, int has_cov) {
+  uint64_t sz = 63ull + spk_len + (has_cov ? 32 : 0);
+  return (sz + 99) / 100;
+}
+
+/* returns 0 ok (mass in *out), -1 incomputable */
+static int calc_storage_mass_tx(const ok_tx *tx, uint64_t *out) {
+  if (ok_tx_is_coinbase(tx)) {
+    *out = 0;
+    return 0;
+  }
+  uint64_t outs_plurality = 0, harmonic_outs = 0;
+  for (uint32_t i = 0; i < tx->n_outputs; i++) {
+    const ok_output *o = &tx->outputs[i];
+    uint64_t p = plurality(o->spk_len, o->cov_id != NULL);
+    if (o->value == 0) return -1; /* zero-value outputs are rejected upstream
+                                     (in-isolation checks); treat as incomputable */
+    uint64_t cp, cpp;
+    if (__builtin_mul_overflow(STORM_PARAM, p, &cp)) return -1;
+    if (__builtin_mul_overflow(cp, p, &cpp)) return -1;
+    uint64_t h;
+    if (__builtin_add_overflow(harmonic_outs, cpp / o->value, &h)) return -1;
+    harmonic_outs = h;
+    outs_plurality += p;
+  }
+  /* relaxed path: |O| = 1, |I| = 1, or |O| = |I| = 2 */
+  int relaxed;
+  uint64_t ins_plurality_small = 0;
+  if (outs_plurality == 1) {
+    relaxed = 1;
+  } else if (tx->n_inputs > 2) {
+    relaxed = 0;
+  } else {
+    for (uint32_t i = 0; i < tx->n_inputs; i++)
+      ins_plurality_small +=
+          plurality(tx->inputs[i].utxo_spk_len, tx->inputs[i].utxo_covenant_id != NULL);
+    relaxed = (ins_plurality_small == 1) ||
+              (outs_plurality == 2 && ins_plurality_small == 2);
+  }
+  if (relaxed) {
+    uint64_t harmonic_ins = 0;
+    for (uint32_t i = 0; i < tx->n_inputs; i++) {
+      const ok_input *in = &tx->inputs[i];
+      uint64_t p = plurality(in->utxo_spk_len, in->utxo_covenant_id != NULL);
+      if (in->utxo_amount == 0) return -1;
+      uint64_t c = STORM_PARAM * p * p / in->utxo_amount; /* no overflow: plurality
+                       bounded by max spk len (verify_utxo_plurality_limits) */
+      uint64_t s = harmonic_ins + c;
+      harmonic_ins = s < harmonic_ins ? UINT64_MAX : s; /* saturating_add */
+    }
+    *out = harmonic_outs > harmonic_ins ? harmonic_outs - harmonic_ins : 0;
+    return 0;
+  }
+  uint64_t ins_plurality = 0, sum_ins = 0;
+  for (uint32_t i = 0; i < tx->n_inputs; i++) {
+    const ok_input *in = &tx->inputs[i];
+    ins_plurality += plurality(in->utxo_spk_len, in->utxo_covenant_id != NULL);
+    sum_ins += in->utxo_amount;
+  }
+  uint64_t mean_ins = sum_ins / ins_plurality;
+  if (mean_ins == 0) mean_ins = 1;
+  uint64_t arith;
+  if (__builtin_mul_overflow(ins_plurality, STORM_PARAM / mean_ins, &arith))
+    arith = UINT64_MAX; /* saturating_mul */
+  *out = harmonic_outs > arith ? harmonic_outs - arith : 0;
+  return 0;
+}
+
 static int validate_one_tx(const ok_tx *tx, uint64_t pov_daa, uint32_t flags,
                            uint64_t coinbase_maturity, uint64_t mass_per_sig_op,
                            uint64_t *fee_out) {
@@ -50,9 +124,12 @@ static int validate_one_tx(const ok_tx *tx, uint64_t pov_daa, uint32_t flags,
   for (uint32_t i = 0; i < tx->n_outputs; i++) total_out += tx->outputs[i].value;
   if (total_in < total_out) return KV_ERR_SPEND_TOO_HIGH;
   uint64_t fee = total_in - total_out;
-  /* 4. mass commitment — round-1: must be skipped explicitly */
-  if (flags != KV_FLAGS_SKIP_MASS_CHECK && flags != KV_FLAGS_SKIP_SCRIPT_CHECKS)
-    return KV_ERR_BAD_BLOB;
+  /* 4. mass commitment (check_mass_commitment, :126-134) */
+  if (flags != KV_FLAGS_SKIP_MASS_CHECK) {
+    uint64_t calc = 0;
+    if (calc_storage_mass_tx(tx, &calc)) return KV_ERR_MASS_INCOMPUTABLE;
+    if (calc != tx->storage_mass) return KV_ERR_WRONG_MASS;
+  }
   /* 5. sequence locks (:136-161) */
   for (uint32_t i = 0; i < tx->n_inputs; i++) {
     const ok_input *in = &tx->inputs[i];

@@ -192,5 +192,44 @@ def gen_block(oracle, seed: int, n_txs: int, *, pct_multi_input=0, pct_ecdsa=0,
                 sig_script = bytes(sig_script)
             txs[t]["inputs"][i]["sig_script"] = sig_script
 
+    # storage-mass commitment (KIP-0009; mirrors mass/mod.rs:385-514) so the
+    # blocks validate under KV_FLAGS_FULL as well
+    for t in txs:
+        t["storage_mass"] = storage_mass(t)
+
     blob = B.build_blob(txs)
     return blob, {"n_txs": n_txs, "specs": specs}
+
+
+STORM = 10**8 * 10**4
+
+
+def _plurality(spk_len, has_cov):
+    return -(-(63 + spk_len + (32 if has_cov else 0)) // 100)
+
+
+def storage_mass(tx):
+    outs_p = 0
+    harm_outs = 0
+    for o in tx["outputs"]:
+        p = _plurality(len(o["spk"]), bool(o["covenant"]))
+        harm_outs += STORM * p * p // o["value"]
+        outs_p += p
+    ins = tx["inputs"]
+    if outs_p == 1:
+        relaxed = True
+    elif len(ins) > 2:
+        relaxed = False
+    else:
+        ins_p = sum(_plurality(len(i["utxo"]["spk"]), bool(i["utxo"]["covenant_id"]))
+                    for i in ins)
+        relaxed = ins_p == 1 or (outs_p == 2 and ins_p == 2)
+    if relaxed:
+        harm_ins = sum(STORM * _plurality(len(i["utxo"]["spk"]),
+                                          bool(i["utxo"]["covenant_id"])) ** 2
+                       // i["utxo"]["amount"] for i in ins)
+        return max(0, harm_outs - harm_ins)
+    ins_p = sum(_plurality(len(i["utxo"]["spk"]), bool(i["utxo"]["covenant_id"]))
+                for i in ins)
+    mean = max(1, sum(i["utxo"]["amount"] for i in ins) // ins_p)
+    return max(0, harm_outs - ins_p * (STORM // mean))

@@ -845,8 +845,14 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
       for (auto &o : tx.outputs) total_out += o.value;
       if (total_in < total_out) code = KV_ERR_SPEND_TOO_HIGH;
     }
-    if (!code && flags != KV_FLAGS_SKIP_MASS_CHECK && flags != KV_FLAGS_SKIP_SCRIPT_CHECKS)
-      code = KV_ERR_BAD_BLOB; /* storage-mass check not yet restated: fail loudly */
+    if (!code && flags != KV_FLAGS_SKIP_MASS_CHECK) {
+      /* check_mass_commitment (tx_validation_in_utxo_context.rs:126-134) */
+      uint64_t calc = 0;
+      if (kvh_storage_mass(tx, false, &calc))
+        code = KV_ERR_MASS_INCOMPUTABLE;
+      else if (calc != tx.storage_mass)
+        code = KV_ERR_WRONG_MASS;
+    }
     if (!code) {
       for (auto &in : tx.inputs) {
         if ((in.sequence & KVH_SEQ_DISABLED) == KVH_SEQ_DISABLED) continue;

@@ -22,34 +22,35 @@ def engine():
     eng.close()
 
 
-def oracle_validate(oracle, blob, n):
+def oracle_validate(oracle, blob, n, flags=SKIP_MASS):
     codes = (ctypes.c_int32 * n)()
     fees = (ctypes.c_uint64 * n)()
     mh = (ctypes.c_uint8 * 32)()
-    rc = oracle.ok_validate_block_parallel(blob, len(blob), 10**9, 10**9, SKIP_MASS,
+    rc = oracle.ok_validate_block_parallel(blob, len(blob), 10**9, 10**9, flags,
                                            8, codes, fees, mh)
     assert rc == 0
     return list(codes), list(fees), bytes(mh)
 
 
-def engine_validate(engine, blob, n):
-    codes, fees, partial = engine.validate_block(blob, n, 10**9, 10**9, SKIP_MASS)
+def engine_validate(engine, blob, n, flags=SKIP_MASS):
+    codes, fees, partial = engine.validate_block(blob, n, 10**9, 10**9, flags)
     mh = engine.muhash_finalize(partial)
     return codes, fees, mh
 
 
-@pytest.mark.parametrize("kwargs", [
-    dict(seed=10, n_txs=64),                                            # config-1 shape
-    dict(seed=11, n_txs=120, pct_multi_input=20, pct_ecdsa=10),         # config-3 shape
-    dict(seed=12, n_txs=80, pct_multi_input=20, pct_ecdsa=10,
-         pct_multisig=10, pct_invalid=15),                              # adversarial
-    dict(seed=13, n_txs=30, payload_len=500),                           # payloads
+@pytest.mark.parametrize("kwargs,flags", [
+    (dict(seed=10, n_txs=64), SKIP_MASS),                               # config-1 shape
+    (dict(seed=11, n_txs=120, pct_multi_input=20, pct_ecdsa=10), SKIP_MASS),  # config-3
+    (dict(seed=12, n_txs=80, pct_multi_input=20, pct_ecdsa=10,
+          pct_multisig=10, pct_invalid=15), SKIP_MASS),                 # adversarial
+    (dict(seed=13, n_txs=30, payload_len=500), SKIP_MASS),              # payloads
+    (dict(seed=14, n_txs=60, pct_multi_input=30, pct_ecdsa=10), 0),     # FULL (KIP-9 mass)
 ])
-def test_validate_block_parity(oracle, engine, kwargs):
+def test_validate_block_parity(oracle, engine, kwargs, flags):
     n = kwargs["n_txs"]
     blob, meta = gen_block(oracle, **kwargs)
-    oc, of, omh = oracle_validate(oracle, blob, n)
-    ec, ef, emh = engine_validate(engine, blob, n)
+    oc, of, omh = oracle_validate(oracle, blob, n, flags)
+    ec, ef, emh = engine_validate(engine, blob, n, flags)
     assert ec == oc, [(i, a, b) for i, (a, b) in enumerate(zip(ec, oc)) if a != b][:5]
     assert ef == of
     assert emh == omh

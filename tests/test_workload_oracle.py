@@ -60,3 +60,28 @@ def test_muhash_changes_with_verdicts(oracle):
     # all txs invalid → muhash over empty set = EMPTY_MUHASH
     assert mh_none.hex() == (
         "544eb3142c000f0ad2c76ac41f4222abbababed830eeafee4b6dc56b52d5cac0")
+
+
+def test_full_flags_storage_mass(oracle):
+    """KV_FLAGS_FULL: the generator's storage-mass commitments must satisfy the
+    KIP-9 check; a corrupted commitment must yield WrongMass (5)."""
+    import ctypes
+    blob, meta = gen_block(oracle, seed=6, n_txs=16, pct_multi_input=30)
+    n = 16
+    codes = (ctypes.c_int32 * n)()
+    fees = (ctypes.c_uint64 * n)()
+    mh = (ctypes.c_uint8 * 32)()
+    rc = oracle.ok_validate_block_parallel(blob, len(blob), 10**9, 10**9, 0, 8,
+                                           codes, fees, mh)
+    assert rc == 0
+    assert all(c == 0 for c in codes), list(codes)
+    # corrupt one tx's committed mass → WrongMass
+    import struct
+    off = struct.unpack("<I", blob[4:8])[0]
+    bad = bytearray(blob)
+    bad[off + 48:off + 56] = struct.pack("<Q", 12345678)
+    rc = oracle.ok_validate_block_parallel(bytes(bad), len(bad), 10**9, 10**9, 0, 8,
+                                           codes, fees, mh)
+    assert rc == 0
+    assert codes[0] == 5  # KV_ERR_WRONG_MASS
+    assert all(c == 0 for c in list(codes)[1:])
