@@ -198,6 +198,21 @@ int kv_stage_tuples(kv_ctx *ctx, const uint8_t *tuples, size_t n, int ecdsa);
 int kv_verify_staged(kv_ctx *ctx, size_t n, int ecdsa, double *kernel_ms);
 int kv_fetch_bitmap(kv_ctx *ctx, size_t n, uint64_t *bitmap_out);
 
+/* ---- GPU-resident UTXO set ----
+ * ⇔ UtxoCollection (consensus/core/src/utxo/utxo_collection.rs:5) + the
+ * populate/diff steps (utxo_validation.rs:351-390, utxo_diff.rs:224).
+ * Outpoints are 36B (tx_id‖index LE); entries are packed 64B records:
+ * amount u64 ‖ daa_score u64 ‖ flags u16 (bit0 coinbase) ‖ spk_version u16 ‖
+ * spk_len u32 ‖ spk[36] inline (standard SPKs ≤ 35B; larger scripts are a
+ * documented round-2 arena extension). */
+int kv_utxo_reset(kv_ctx *ctx, uint64_t capacity);
+int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints, const uint8_t *entries64,
+                   size_t n);
+int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n);
+int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
+                   uint8_t *entries_out /*[64n] or NULL*/, uint64_t *found_bitmap,
+                   double *kernel_ms /*optional*/);
+
 /* Sig-cache statistics (crypto/txscript/src/caches.rs:57-82 counters). */
 typedef struct {
   uint64_t insertions;

@@ -22,6 +22,7 @@
  * hipcc invocation, no -fgpu-rdc) */
 #include "kv_secp_kernels.hip"
 #include "kv_sighash_kernels.hip"
+#include "kv_utxo_kernels.hip"
 #include "kv_validate_host.inc"
 
 static thread_local std::string g_last_error;
@@ -51,6 +52,15 @@ struct kv_ctx {
   uint8_t *d_status = nullptr;
   size_t d_status_cap = 0;
   uint64_t cache_hits = 0, cache_misses = 0, cache_insertions = 0;
+  /* GPU-resident UTXO set */
+  kv::utxo_slot *d_utxo = nullptr;
+  uint64_t utxo_cap = 0; /* power of two */
+  uint8_t *d_op_in = nullptr;
+  size_t d_op_cap = 0;
+  uint8_t *d_val_in = nullptr;
+  size_t d_val_cap = 0;
+  uint8_t *d_ent_out = nullptr;
+  size_t d_ent_cap = 0;
 };
 
 static int ensure_cap(void **ptr, size_t *cap, size_t need) {
@@ -1019,5 +1029,122 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
     if (codes[t]) fees[t] = 0; /* fee defined only for accepted txs */
   memcpy(tx_codes_out, codes.data(), (size_t)n_txs * 4);
   memcpy(fees_out, fees.data(), (size_t)n_txs * 8);
+  return 0;
+}
+
+
+/* ---------------- GPU-resident UTXO set (kv_utxo_kernels.hip) ----------------
+ * ⇔ utxo_collection.rs:5 HashMap + the populate/diff steps
+ * (utxo_validation.rs:351-390, utxo_diff.rs:224). Entries are packed 64B
+ * records: amount u64 ‖ daa u64 ‖ flags u16 (bit0 coinbase) ‖ spk_version u16 ‖
+ * spk_len u32 ‖ spk[36] inline. */
+
+extern "C" int kv_utxo_reset(kv_ctx *ctx, uint64_t capacity) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  uint64_t cap = 64;
+  while (cap < capacity * 2) cap <<= 1; /* ≤50% load factor */
+  if (ctx->d_utxo) (void)hipFree(ctx->d_utxo);
+  if (hipMalloc(&ctx->d_utxo, cap * sizeof(kv::utxo_slot)) != hipSuccess) {
+    ctx->d_utxo = nullptr;
+    ctx->utxo_cap = 0;
+    set_error("kv_utxo_reset: hipMalloc failed");
+    return -2;
+  }
+  HIP_CHECK(hipMemsetAsync(ctx->d_utxo, 0, cap * sizeof(kv::utxo_slot), ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  ctx->utxo_cap = cap;
+  return 0;
+}
+
+static int utxo_stage(kv_ctx *ctx, const uint8_t *outpoints, const uint8_t *values,
+                      size_t n) {
+  if (ensure_cap((void **)&ctx->d_op_in, &ctx->d_op_cap, n * 36)) return -2;
+  HIP_CHECK(hipMemcpyAsync(ctx->d_op_in, outpoints, n * 36, hipMemcpyHostToDevice,
+                           ctx->stream));
+  if (values) {
+    if (ensure_cap((void **)&ctx->d_val_in, &ctx->d_val_cap, n * 64)) return -2;
+    HIP_CHECK(hipMemcpyAsync(ctx->d_val_in, values, n * 64, hipMemcpyHostToDevice,
+                             ctx->stream));
+  }
+  return 0;
+}
+
+extern "C" int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints,
+                              const uint8_t *entries64, size_t n) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  if (!ctx->d_utxo) {
+    set_error("kv_utxo_upsert: call kv_utxo_reset first");
+    return -1;
+  }
+  int rc = utxo_stage(ctx, outpoints, entries64, n);
+  if (rc) return rc;
+  static int *d_fail = nullptr;
+  if (!d_fail) HIP_CHECK(hipMalloc(&d_fail, 4));
+  HIP_CHECK(hipMemsetAsync(d_fail, 0, 4, ctx->stream));
+  hipLaunchKernelGGL(kv::kv_utxo_upsert_kernel, dim3(((uint32_t)n + 255) / 256),
+                     dim3(256), 0, ctx->stream, ctx->d_utxo, ctx->utxo_cap - 1,
+                     ctx->d_op_in, ctx->d_val_in, (unsigned long long)n, d_fail);
+  int fail = 0;
+  HIP_CHECK(hipMemcpyAsync(&fail, d_fail, 4, hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  if (fail) {
+    set_error("kv_utxo_upsert: table full");
+    return -3;
+  }
+  return 0;
+}
+
+extern "C" int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  if (!ctx->d_utxo) {
+    set_error("kv_utxo_remove: call kv_utxo_reset first");
+    return -1;
+  }
+  int rc = utxo_stage(ctx, outpoints, nullptr, n);
+  if (rc) return rc;
+  hipLaunchKernelGGL(kv::kv_utxo_remove_kernel, dim3(((uint32_t)n + 255) / 256),
+                     dim3(256), 0, ctx->stream, ctx->d_utxo, ctx->utxo_cap - 1,
+                     ctx->d_op_in, (unsigned long long)n);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return 0;
+}
+
+extern "C" int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
+                              uint8_t *entries_out, uint64_t *found_bitmap,
+                              double *kernel_ms) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  if (!ctx->d_utxo) {
+    set_error("kv_utxo_lookup: call kv_utxo_reset first");
+    return -1;
+  }
+  int rc = utxo_stage(ctx, outpoints, nullptr, n);
+  if (rc) return rc;
+  size_t words = (n + 63) / 64;
+  if (ensure_cap((void **)&ctx->d_ent_out, &ctx->d_ent_cap, n * 64)) return -2;
+  if (ensure_cap((void **)&ctx->d_bitmap, &ctx->d_bitmap_cap, words * 8)) return -2;
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventRecord(t0, ctx->stream));
+  hipLaunchKernelGGL(kv::kv_utxo_lookup_kernel, dim3(((uint32_t)n + 255) / 256),
+                     dim3(256), 0, ctx->stream, ctx->d_utxo, ctx->utxo_cap - 1,
+                     ctx->d_op_in, (unsigned long long)n, ctx->d_ent_out,
+                     (unsigned long long *)ctx->d_bitmap);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipEventRecord(t1, ctx->stream));
+  if (entries_out)
+    HIP_CHECK(hipMemcpyAsync(entries_out, ctx->d_ent_out, n * 64,
+                             hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(found_bitmap, ctx->d_bitmap, words * 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  if (kernel_ms) {
+    float ms = 0.f;
+    HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+    *kernel_ms = ms;
+  }
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
   return 0;
 }
