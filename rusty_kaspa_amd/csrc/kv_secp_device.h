@@ -34,6 +34,9 @@ namespace kv {
 
 typedef uint64_t u64;
 typedef uint32_t u32;
+typedef unsigned __int128 u128; /* 128-bit accumulators: LLVM lowers their
+  adds to native v_add_co/v_addc carry chains — the hand-rolled (s<a) carry
+  extraction compiled to v_cmp+v_cndmask chains, ~2× the instructions */
 
 /* ---------- 256-bit field element mod p = 2^256 - 0x1000003D1 ---------- */
 
@@ -46,19 +49,15 @@ struct fe {
 #define KV_PC 0x1000003D1ULL /* 2^256 - p */
 
 __device__ __forceinline__ u64 addc(u64 a, u64 b, u64 &carry) {
-  u64 s = a + b;
-  u64 c1 = s < a;
-  u64 s2 = s + carry;
-  carry = c1 + (s2 < s);
-  return s2;
+  u128 t = (u128)a + b + carry;
+  carry = (u64)(t >> 64);
+  return (u64)t;
 }
 
 __device__ __forceinline__ u64 subb(u64 a, u64 b, u64 &borrow) {
-  u64 d = a - b;
-  u64 b1 = a < b;
-  u64 d2 = d - borrow;
-  borrow = b1 + (d < borrow);
-  return d2;
+  u128 t = (u128)a - b - borrow;
+  borrow = (u64)(t >> 64) & 1;
+  return (u64)t;
 }
 
 __device__ __forceinline__ int fe_gte_p(const fe &a) {
@@ -91,25 +90,31 @@ __device__ __forceinline__ void fe_norm_once(fe &a) {
 }
 
 __device__ __forceinline__ void fe_add(fe &r, const fe &a, const fe &b) {
-  u64 carry = 0;
-  r.n[0] = addc(a.n[0], b.n[0], carry);
-  r.n[1] = addc(a.n[1], b.n[1], carry);
-  r.n[2] = addc(a.n[2], b.n[2], carry);
-  r.n[3] = addc(a.n[3], b.n[3], carry);
-  /* wrap: += carry * PC (carry 0/1) */
-  u64 add0 = carry * KV_PC;
-  u64 c2 = 0;
-  r.n[0] = addc(r.n[0], add0, c2);
-  r.n[1] = addc(r.n[1], 0, c2);
-  r.n[2] = addc(r.n[2], 0, c2);
-  r.n[3] = addc(r.n[3], 0, c2);
-  /* c2 can only be set if r wrapped again with tiny value; fold once more */
-  u64 add1 = c2 * KV_PC;
-  u64 c3 = 0;
-  r.n[0] = addc(r.n[0], add1, c3);
-  r.n[1] = addc(r.n[1], 0, c3);
-  r.n[2] = addc(r.n[2], 0, c3);
-  r.n[3] = addc(r.n[3], 0, c3);
+  u128 c = (u128)a.n[0] + b.n[0];
+  u64 t0 = (u64)c;
+  c = (c >> 64) + a.n[1] + b.n[1];
+  u64 t1 = (u64)c;
+  c = (c >> 64) + a.n[2] + b.n[2];
+  u64 t2 = (u64)c;
+  c = (c >> 64) + a.n[3] + b.n[3];
+  u64 t3 = (u64)c;
+  /* wrap 2^256 → += PC; a+b < 2^257 so the second wrap adds < 2^34: no ripple
+   * past limb 1 is possible after the first fold unless limbs были max —
+   * handled by a full u128 chain again */
+  u64 w = (u64)(c >> 64) * KV_PC;
+  c = (u128)t0 + w;
+  r.n[0] = (u64)c;
+  c = (c >> 64) + t1;
+  r.n[1] = (u64)c;
+  c = (c >> 64) + t2;
+  r.n[2] = (u64)c;
+  c = (c >> 64) + t3;
+  r.n[3] = (u64)c;
+  if ((u64)(c >> 64)) { /* wrapped again: value is tiny */
+    c = (u128)r.n[0] + KV_PC;
+    r.n[0] = (u64)c;
+    r.n[1] += (u64)(c >> 64);
+  }
   fe_norm_once(r);
 }
 
@@ -144,13 +149,9 @@ __device__ __forceinline__ void fe_mul_inner(u64 t[8], const u64 *an, const u64 
     u64 carry = 0;
 #pragma unroll
     for (int j = 0; j < 4; j++) {
-      u64 lo = an[i] * bn[j];
-      u64 hi = __umul64hi(an[i], bn[j]);
-      u64 c = 0;
-      t[i + j] = addc(t[i + j], lo, c);
-      u64 c2 = 0;
-      t[i + j] = addc(t[i + j], carry, c2);
-      carry = hi + c + c2; /* hi <= 2^64-2: no overflow */
+      u128 cur = (u128)an[i] * bn[j] + t[i + j] + carry;
+      t[i + j] = (u64)cur;
+      carry = (u64)(cur >> 64);
     }
     t[i + 4] = carry;
   }
@@ -158,34 +159,32 @@ __device__ __forceinline__ void fe_mul_inner(u64 t[8], const u64 *an, const u64 
 
 __device__ __forceinline__ void fe_reduce8(fe &r, const u64 t[8]) {
   /* fold 1: s[0..4] = t[0..3] + PC * t[4..7] (PC is 33 bits) */
-  u64 lo0 = t[4] * KV_PC, hi0 = __umul64hi(t[4], KV_PC);
-  u64 lo1 = t[5] * KV_PC, hi1 = __umul64hi(t[5], KV_PC);
-  u64 lo2 = t[6] * KV_PC, hi2 = __umul64hi(t[6], KV_PC);
-  u64 lo3 = t[7] * KV_PC, hi3 = __umul64hi(t[7], KV_PC);
-  u64 cA = 0, cB = 0;
-  u64 s0 = addc(t[0], lo0, cA);
-  u64 s1 = addc(t[1], lo1, cA);
-  u64 s2 = addc(t[2], lo2, cA);
-  u64 s3 = addc(t[3], lo3, cA);
-  u64 s4 = cA;
-  s1 = addc(s1, hi0, cB);
-  s2 = addc(s2, hi1, cB);
-  s3 = addc(s3, hi2, cB);
-  s4 = s4 + hi3 + cB; /* < 2^34 */
-  /* fold 2: r = s[0..3] + PC * s4 */
-  u64 lo = s4 * KV_PC, hi = __umul64hi(s4, KV_PC);
-  u64 c = 0;
-  r.n[0] = addc(s0, lo, c);
-  r.n[1] = addc(s1, hi, c);
-  r.n[2] = addc(s2, 0, c);
-  r.n[3] = addc(s3, 0, c);
-  /* c set → value wrapped 2^256 exactly once and is now tiny (< 2^68) */
-  u64 add1 = c * KV_PC;
-  u64 c2 = 0;
-  r.n[0] = addc(r.n[0], add1, c2);
-  r.n[1] = addc(r.n[1], 0, c2);
-  r.n[2] = addc(r.n[2], 0, c2);
-  r.n[3] = addc(r.n[3], 0, c2);
+  u128 c = (u128)t[0] + (u128)t[4] * KV_PC;
+  u64 s0 = (u64)c;
+  c = (c >> 64) + t[1] + (u128)t[5] * KV_PC;
+  u64 s1 = (u64)c;
+  c = (c >> 64) + t[2] + (u128)t[6] * KV_PC;
+  u64 s2 = (u64)c;
+  c = (c >> 64) + t[3] + (u128)t[7] * KV_PC;
+  u64 s3 = (u64)c;
+  u64 s4 = (u64)(c >> 64); /* < 2^34 */
+  /* fold 2: r = s[0..3] + PC * s4 (product < 2^67) */
+  c = (u128)s0 + (u128)s4 * KV_PC;
+  r.n[0] = (u64)c;
+  c = (c >> 64) + s1;
+  r.n[1] = (u64)c;
+  c = (c >> 64) + s2;
+  r.n[2] = (u64)c;
+  c = (c >> 64) + s3;
+  r.n[3] = (u64)c;
+  /* carry out → wrapped 2^256 once more; value tiny (< 2^68) */
+  if ((u64)(c >> 64)) {
+    c = (u128)r.n[0] + KV_PC;
+    r.n[0] = (u64)c;
+    u64 cc = (u64)(c >> 64);
+    r.n[1] += cc;
+    /* (r.n[1] was tiny: no further carry) */
+  }
   fe_norm_once(r);
 }
 
@@ -198,31 +197,29 @@ __device__ __forceinline__ void fe_mul(fe &r, const fe &a, const fe &b) {
 __device__ __forceinline__ void fe_sqr(fe &r, const fe &a) { fe_mul(r, a, a); }
 
 __device__ __forceinline__ void fe_mul_small(fe &r, const fe &a, u64 k) {
-  u64 c = 0, t4;
-  u64 lo0 = a.n[0] * k, hi0 = __umul64hi(a.n[0], k);
-  u64 lo1 = a.n[1] * k, hi1 = __umul64hi(a.n[1], k);
-  u64 lo2 = a.n[2] * k, hi2 = __umul64hi(a.n[2], k);
-  u64 lo3 = a.n[3] * k, hi3 = __umul64hi(a.n[3], k);
-  fe s;
-  s.n[0] = lo0;
-  s.n[1] = addc(lo1, hi0, c);
-  s.n[2] = addc(lo2, hi1, c);
-  s.n[3] = addc(lo3, hi2, c);
-  t4 = hi3 + c;
-  u64 lo = t4 * KV_PC, hi = __umul64hi(t4, KV_PC);
-  c = 0;
-  s.n[0] = addc(s.n[0], lo, c);
-  s.n[1] = addc(s.n[1], hi, c);
-  s.n[2] = addc(s.n[2], 0, c);
-  s.n[3] = addc(s.n[3], 0, c);
-  u64 add1 = c * KV_PC;
-  u64 c2 = 0;
-  s.n[0] = addc(s.n[0], add1, c2);
-  s.n[1] = addc(s.n[1], 0, c2);
-  s.n[2] = addc(s.n[2], 0, c2);
-  s.n[3] = addc(s.n[3], 0, c2);
-  fe_norm_once(s);
-  r = s;
+  u128 c = (u128)a.n[0] * k;
+  u64 t0 = (u64)c;
+  c = (c >> 64) + (u128)a.n[1] * k;
+  u64 t1 = (u64)c;
+  c = (c >> 64) + (u128)a.n[2] * k;
+  u64 t2 = (u64)c;
+  c = (c >> 64) + (u128)a.n[3] * k;
+  u64 t3 = (u64)c;
+  u64 t4 = (u64)(c >> 64);
+  c = (u128)t0 + (u128)t4 * KV_PC;
+  r.n[0] = (u64)c;
+  c = (c >> 64) + t1;
+  r.n[1] = (u64)c;
+  c = (c >> 64) + t2;
+  r.n[2] = (u64)c;
+  c = (c >> 64) + t3;
+  r.n[3] = (u64)c;
+  if ((u64)(c >> 64)) {
+    c = (u128)r.n[0] + KV_PC;
+    r.n[0] = (u64)c;
+    r.n[1] += (u64)(c >> 64);
+  }
+  fe_norm_once(r);
 }
 
 __device__ __forceinline__ int fe_is_zero(const fe &a) {
