@@ -79,6 +79,78 @@ def cpu_baseline_leg(oracle, tuples, n):
     }
 
 
+def block_mode(args):
+    """Secondary metric: txs-validated/sec on the full block path (BASELINE
+    config 3 shape: 300 txs/block, 70% 1-in P2PK schnorr, 20% multi-input,
+    10% ECDSA). One step = one kv_validate_block call over a mergeset-sized
+    batch of blocks (blob handed over as host memory — the real call pattern;
+    the rate is PCIe-inclusive) + the muhash finalize."""
+    import ctypes
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    from workload import gen_block
+    oracle = load_oracle()
+    n_txs = args.block_batch * 300
+    t0 = time.time()
+    blob, _ = gen_block(oracle, seed=SEED, n_txs=n_txs, pct_multi_input=20,
+                        pct_ecdsa=10)
+    log(f"generated {n_txs} mixed txs ({len(blob)/1e6:.1f} MB blob) "
+        f"in {time.time()-t0:.1f}s")
+    from rusty_kaspa_amd.engine import Engine
+    eng = Engine()
+    n_inputs = sum(1 for _ in range(0))  # informational only
+
+    def one_step():
+        codes, fees, partial = eng.validate_block(blob, n_txs, 10**9, 10**9, 0)
+        mh = eng.muhash_finalize(partial)
+        return codes, mh
+
+    for _ in range(args.warmup):
+        codes, _ = one_step()
+    assert all(c == 0 for c in codes), "unexpected invalid txs in bench batch"
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    elapsed = time.perf_counter() - t0
+    value = n_txs * args.steps / elapsed
+
+    cpu_baseline = None
+    if not args.skip_cpu_baseline:
+        cores = os.cpu_count() or 8
+        codes_a = (ctypes.c_int32 * n_txs)()
+        fees_a = (ctypes.c_uint64 * n_txs)()
+        mh_a = (ctypes.c_uint8 * 32)()
+        t0 = time.perf_counter()
+        oracle.ok_validate_block_parallel(blob, len(blob), 10**9, 10**9, 0, cores,
+                                          codes_a, fees_a, mh_a)
+        dt = time.perf_counter() - t0
+        cpu_baseline = {"value": round(n_txs / dt, 1), "unit": "txs-validated/sec",
+                        "cores": cores, "kind": "port",
+                        "sample": f"{n_txs} txs, one full pass, {dt:.1f}s wall"}
+
+    result = {
+        "metric": "txs-validated/sec",
+        "value": round(value, 1),
+        "unit": "txs/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1000, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "u256",
+        "data": "synthetic (seeded oracle-signed config-3 mix; batch reused across steps)",
+        "config": {"workload": "block-validate-config3",
+                   "blocks_per_step": args.block_batch, "txs_per_block": 300,
+                   "mix": "70p2pk/20multi-in/10ecdsa", "flags": "FULL",
+                   "parallelism": "single"},
+        "roofline": None,  # per-kernel rooflines live in the default mode + profiles/
+        "cpu_baseline": cpu_baseline,
+    }
+    print(json.dumps(result), flush=True)
+    eng.close()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -87,7 +159,13 @@ def main():
     ap.add_argument("--tuples", type=int, default=DEFAULT_TUPLES)
     ap.add_argument("--invalid-permille", type=int, default=INVALID_PERMILLE)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--mode", choices=["verify", "block"], default="verify")
+    ap.add_argument("--block-batch", type=int, default=16,
+                    help="blocks (of 300 txs) per step in --mode block")
     args = ap.parse_args()
+
+    if args.mode == "block":
+        return block_mode(args)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
