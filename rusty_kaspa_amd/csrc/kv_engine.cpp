@@ -1010,13 +1010,13 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
         uint32_t n_cur = (uint32_t)cnt;
         uint64_t *cur = src;
         while (n_cur > 1) {
-          /* halving keeps every pass massively parallel (1 mulmod per lane);
-           * a per-thread mulmod is scratch-latency bound (~0.4 ms), so deep
-           * chains on few threads lose badly. Wave-cooperative mulmod is the
-           * planned fix for the tail passes. */
-          uint32_t stride = n_cur > 2048 ? 1024 : (n_cur + 1) / 2;
-          hipLaunchKernelGGL(kv::kv_u3072_reduce_kernel, dim3((stride + 255) / 256),
-                             dim3(256), 0, ctx->stream, cur, n_cur, stride, pa);
+          /* wave-cooperative mulmod (one value per wave, limbs in registers):
+           * two passes — wide fan-in then a single wave over ≤512 partials */
+          uint32_t stride = n_cur > 512 ? 512 : 1;
+          uint32_t waves_per_block = 4; /* 256 threads */
+          uint32_t blocks = (stride + waves_per_block - 1) / waves_per_block;
+          hipLaunchKernelGGL(kv_u3072_reduce_wave_kernel, dim3(blocks), dim3(256),
+                             0, ctx->stream, cur, n_cur, stride, pa);
           n_cur = stride;
           cur = pa;
           std::swap(pa, pb);

@@ -329,3 +329,85 @@ extern "C" __global__ void kv_u3072_reduce_kernel(const uint64_t *__restrict__ e
 }
 
 } // namespace kv
+
+/* ---------------- wave-cooperative U3072 mulmod ----------------
+ * One 3072-bit value per WAVE: lane l (l < 48) holds limb l in a register —
+ * no scratch at all (the per-thread mulmod above is scratch-latency bound).
+ * Column k of a·b mod (2^3072 − C):
+ *   S_k = Σ_i a_i · b_{(k−i) mod 48} · (C if i > k else 1)
+ * accumulated per lane in a 192-bit (lo,mid,hi) register triple via cross-lane
+ * shfl broadcasts, then redistributed (mid→k+1, hi→k+2, wrapping ×C) and
+ * carry-propagated lane-to-lane until quiescent. CDNA4-native: wave64 shfl is
+ * ds_bpermute, wholly in-register. */
+
+__device__ inline uint64_t u3072_wave_mulmod(uint64_t a_limb, uint64_t b_limb) {
+  const uint64_t C = KVU_PRIME_DIFF;
+  int lane = threadIdx.x & 63;
+  int k = lane < 48 ? lane : 0;
+  uint64_t lo = 0, mid = 0, hi = 0;
+#pragma unroll 1
+  for (int i = 0; i < 48; i++) {
+    uint64_t ai = __shfl((long long)a_limb, i);
+    int j = k - i;
+    uint64_t w = 1;
+    if (j < 0) {
+      j += 48;
+      w = C;
+    }
+    uint64_t bj = __shfl((long long)b_limb, j);
+    unsigned __int128 p = (unsigned __int128)ai * bj;
+    uint64_t plo = (uint64_t)p, phi = (uint64_t)(p >> 64);
+    unsigned __int128 t = (unsigned __int128)plo * w;   /* ≤ 2^85 when w=C */
+    unsigned __int128 t2 = (unsigned __int128)phi * w;
+    unsigned __int128 c = (unsigned __int128)lo + (uint64_t)t;
+    lo = (uint64_t)c;
+    c = (c >> 64) + (uint64_t)(t >> 64) + (unsigned __int128)mid + (uint64_t)t2;
+    mid = (uint64_t)c;
+    hi += (uint64_t)(c >> 64) + (uint64_t)(t2 >> 64);
+  }
+  /* redistribute mid→column k+1, hi→column k+2 (wrap ×C past column 47) */
+  uint64_t mid_in = __shfl((long long)mid, (k + 47) % 48);
+  uint64_t hi_in = __shfl((long long)hi, (k + 46) % 48);
+  uint64_t w1 = (k == 0) ? C : 1;
+  uint64_t w2 = (k <= 1) ? C : 1;
+  unsigned __int128 s = (unsigned __int128)lo + (unsigned __int128)mid_in * w1 +
+                        (unsigned __int128)hi_in * w2;
+  uint64_t limb = (uint64_t)s;
+  uint64_t carry = (uint64_t)(s >> 64);
+  /* lane-to-lane carry propagation until quiescent (values shrink fast) */
+  while (__any(lane < 48 && carry != 0)) {
+    uint64_t cin = __shfl((long long)carry, (k + 47) % 48);
+    if (k == 0) {
+      unsigned __int128 cc = (unsigned __int128)cin * C;
+      s = (unsigned __int128)limb + (uint64_t)cc;
+      carry = (uint64_t)(s >> 64) + (uint64_t)(cc >> 64);
+    } else {
+      s = (unsigned __int128)limb + cin;
+      carry = (uint64_t)(s >> 64);
+    }
+    limb = (uint64_t)s;
+  }
+  return limb;
+}
+
+/* one WAVE per partial: wave w chains elements w, w+stride, … */
+extern "C" __global__ void kv_u3072_reduce_wave_kernel(
+    const uint64_t *__restrict__ elements, uint32_t n, uint32_t stride,
+    uint64_t *__restrict__ partials) {
+  uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) / 64;
+  int lane = threadIdx.x & 63;
+  if (wave >= stride) return;
+  uint64_t acc = 0;
+  int any = 0;
+  for (uint32_t i = wave; i < n; i += stride) {
+    uint64_t e = lane < 48 ? elements[(size_t)i * KVU_LIMBS + lane] : 0;
+    if (!any) {
+      acc = e;
+      any = 1;
+    } else {
+      acc = u3072_wave_mulmod(acc, e);
+    }
+  }
+  if (!any) acc = (lane == 0) ? 1 : 0; /* identity */
+  if (lane < 48) partials[(size_t)wave * KVU_LIMBS + lane] = acc;
+}
