@@ -47,8 +47,15 @@ def log(msg):
 
 
 def load_oracle():
-    subprocess.run(["make", "-s", "-C", os.path.join(REPO, "oracle")], check=True)
-    return ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
+    so = os.path.join(REPO, "oracle", "liboracle.so")
+    if not os.path.exists(so) and int(os.environ.get("LOCAL_RANK", "0")) == 0:
+        # only rank 0 builds; the .so normally ships prebuilt in-tree
+        subprocess.run(["make", "-s", "-C", os.path.join(REPO, "oracle")], check=True)
+    for _ in range(60):
+        if os.path.exists(so):
+            break
+        time.sleep(1)
+    return ctypes.CDLL(so)
 
 
 def gen_tuples(oracle, n, seed, invalid_permille):

@@ -891,11 +891,13 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
   /* phase 2: GPU — subhashes, sighash+tuple assembly, EC verify */
   size_t ns = sjobs.size(), ne = ejobs.size();
   std::vector<uint8_t> s_status(ns), e_status(ne);
+  bool blob_uploaded = false;
   if (ns + ne > 0) {
     if (g_vb.blob.ensure(blob_len) || g_vb.subhashes.ensure((size_t)n_txs * 160))
       return -2;
     HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
                              ctx->stream));
+    blob_uploaded = true;
     hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256), dim3(256),
                        0, ctx->stream, (const uint8_t *)g_vb.blob.p, (uint32_t)n_txs,
                        (uint8_t *)g_vb.subhashes.p);
@@ -987,8 +989,9 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
           g_vb.partials_a.ensure(1024 * KVU_LIMBS * 8) ||
           g_vb.partials_b.ensure(1024 * KVU_LIMBS * 8))
         return -2;
-      HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
-                               ctx->stream));
+      if (!blob_uploaded) /* phase 2 may have uploaded it already */
+        HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
+                                 ctx->stream));
       HIP_CHECK(hipMemcpyAsync(g_vb.elem_jobs.p, all.data(),
                                n_all * sizeof(kv::kv_elem_job), hipMemcpyHostToDevice,
                                ctx->stream));
