@@ -17,7 +17,7 @@ for t in range(N):
     inp = B.tx_input(pid, t & 3, sig_script=bytes([0x41]) + bytes(65),
                      commit_kind=0, commit_value=1,
                      utxo=B.utxo_entry(10_000 + t, spk, 42))
-    txs.append(B.tx_dict(0, [inp], [B.tx_output(9_000, spk)], tx_id=pid + bytes(24)))
+    txs.append(B.tx_dict(0, [inp], [B.tx_output(9_000, spk)], tx_id=pid))
 blob = B.build_blob(txs)
 print(f"blob: {len(blob)/1e6:.1f} MB, {N} txs/inputs", flush=True)
 
