@@ -121,39 +121,192 @@ extern "C" __global__ void kv_ec_table_init_kernel() {
   }
 }
 
-/* R = gs*G + ps*P — 4-bit fixed-window interleaved double-and-add.
- * Per 4-bit window: 4 doublings + one mixed add against the shared affine
- * G table + one full Jacobian add against the per-lane P table, each selected
- * with cmov (digit 0 discards the add) so the wave stays converged.
- * ≈ 256 doubles + 64 mixed + 64 full adds + 15-entry P-table build —
- * ~1.8× fewer field multiplies than the per-bit form it replaces. */
+/* ---------------- GLV endomorphism scalar decomposition ----------------
+ * secp256k1 has an efficient endomorphism φ(x,y) = (β·x, y) with φ(P) = λ·P.
+ * Splitting each 256-bit scalar k into k1 + k2·λ with |k1|,|k2| ≤ 2^129 halves
+ * the doubling ladder (33 4-bit windows instead of 64). All constants below
+ * were DERIVED in-repo (cube roots of unity via Tonelli–Shanks, lattice basis
+ * via extended Euclid) and verified against φ(G) = λ·G — see the derivation
+ * script output quoted in DESIGN.md. Decomposition uses the shift method:
+ * c_i = (g_i·k) >> 384, k1 = k − c1·a1 − c2·a2, k2 = −(c1·b1 + c2·b2),
+ * computed in 256-bit two's complement. */
+
+__device__ __constant__ static const u64 GLV_BETA[4] = {
+    0xc1396c28719501eeULL, 0x9cf0497512f58995ULL, 0x6e64479eac3434e9ULL,
+    0x7ae96a2b657c0710ULL};
+__device__ __constant__ static const u64 GLV_G1[4] = { /* round(2^384·b2/n) */
+    0xe893209a45dbb031ULL, 0x3daa8a1471e8ca7fULL, 0xe86c90e49284eb15ULL,
+    0x3086d221a7d46bcdULL};
+__device__ __constant__ static const u64 GLV_G2[4] = { /* round(2^384·(−b1)/n) */
+    0x1571b4ae8ac47f71ULL, 0x221208ac9df506c6ULL, 0x6f547fa90abfe4c4ULL,
+    0xe4437ed6010e8828ULL};
+__device__ __constant__ static const u64 GLV_A1[4] = {
+    0xe86c90e49284eb15ULL, 0x3086d221a7d46bcdULL, 0, 0};
+__device__ __constant__ static const u64 GLV_B1[4] = { /* −b1 as two's compl. */
+    0x90ab8056f5401b3dULL, 0x1bbc8129fef177d7ULL, 0xffffffffffffffffULL,
+    0xffffffffffffffffULL};
+__device__ __constant__ static const u64 GLV_A2[4] = {
+    0x657c1108d9d44cfdULL, 0x14ca50f7a8e2f3f6ULL, 1, 0};
+__device__ __constant__ static const u64 GLV_B2[4] = {
+    0xe86c90e49284eb15ULL, 0x3086d221a7d46bcdULL, 0, 0};
+
+/* low-256 product r = (a*b) mod 2^256 (two's-complement arithmetic) */
+__device__ __forceinline__ void mul_low256(u64 r[4], const u64 a[4], const u64 b[4]) {
+  u64 t[8];
+  fe_mul_inner(t, a, b);
+#pragma unroll
+  for (int i = 0; i < 4; i++) r[i] = t[i];
+}
+
+/* c = (g*k) >> 384, a 128-bit result */
+__device__ __forceinline__ void mul_shift384(u64 c[2], const u64 g[4], const u64 k[4]) {
+  u64 t[8];
+  fe_mul_inner(t, g, k);
+  c[0] = t[6];
+  c[1] = t[7];
+}
+
+struct glv_half {
+  u64 d[3];  /* |k_i|, ≤ 2^129 */
+  u64 neg;   /* 1 if k_i < 0 */
+};
+
+__device__ inline void glv_split(const sc &k, glv_half &h1, glv_half &h2) {
+  u64 c1[4] = {0, 0, 0, 0}, c2[4] = {0, 0, 0, 0};
+  mul_shift384(c1, GLV_G1, k.d);
+  mul_shift384(c2, GLV_G2, k.d);
+  u64 t1[4], t2[4], k1[4], k2[4];
+  /* k1 = k − c1·a1 − c2·a2 (mod 2^256) */
+  mul_low256(t1, c1, GLV_A1);
+  mul_low256(t2, c2, GLV_A2);
+  {
+    u64 borrow = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++) k1[i] = subb(k.d[i], t1[i], borrow);
+    borrow = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++) k1[i] = subb(k1[i], t2[i], borrow);
+  }
+  /* k2 = −(c1·b1_neg·(−1) …) — with GLV_B1 = −b1: k2 = c1·(−b1) − c2·b2 */
+  mul_low256(t1, c1, GLV_B1);
+  mul_low256(t2, c2, GLV_B2);
+  {
+    u64 borrow = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++) k2[i] = subb(t1[i], t2[i], borrow);
+  }
+  /* two's-complement abs */
+  u64 n1 = k1[3] >> 63, n2 = k2[3] >> 63;
+  u64 carry = 1;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    u64 v = n1 ? ~k1[i] : k1[i];
+    u128 s = (u128)v + (n1 ? carry : 0);
+    if (n1) {
+      k1[i] = (u64)s;
+      carry = (u64)(s >> 64);
+    } else {
+      k1[i] = v;
+    }
+  }
+  carry = 1;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    u64 v = n2 ? ~k2[i] : k2[i];
+    u128 s = (u128)v + (n2 ? carry : 0);
+    if (n2) {
+      k2[i] = (u64)s;
+      carry = (u64)(s >> 64);
+    } else {
+      k2[i] = v;
+    }
+  }
+  h1.d[0] = k1[0]; h1.d[1] = k1[1]; h1.d[2] = k1[2]; h1.neg = n1;
+  h2.d[0] = k2[0]; h2.d[1] = k2[1]; h2.d[2] = k2[2]; h2.neg = n2;
+}
+
+__device__ __forceinline__ u64 glv_digit(const glv_half &h, int w) {
+  int bit = w * 4;
+  int limb = bit >> 6, sh = bit & 63;
+  u64 v = h.d[limb] >> sh;
+  if (sh > 60 && limb < 2) v |= h.d[limb + 1] << (64 - sh);
+  return v & 15;
+}
+
+/* R = gs·G + ps·P via GLV-split 4-bit windows: 33 window steps of 4 doublings
+ * + 4 selected adds (G, φG, P, φP streams; φ applied at add time as one β·x
+ * field multiply; negative half-scalars negate the added point's y). */
 __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const ge &P,
                                      volatile int *progress = nullptr) {
-  /* P table (Jacobian): ptab[k] = k·P for k = 1..15 (mixed adds) */
+  glv_half g1h, g2h, p1h, p2h;
+  glv_split(gs, g1h, g2h);
+  glv_split(ps, p1h, p2h);
+  /* per-lane P table (Jacobian): ptab[k] = k·P, k = 1..15 */
   gej ptab[16];
   ptab[1].x = P.x;
   ptab[1].y = P.y;
   ptab[1].z = {{1, 0, 0, 0}};
 #pragma unroll 1
   for (int k = 2; k <= 15; k++) gej_add_ge(ptab[k], ptab[k - 1], P);
-  gej_set_infinity(ptab[0]); /* never selected; keeps reads defined */
+  gej_set_infinity(ptab[0]);
+  fe beta;
+#pragma unroll
+  for (int i = 0; i < 4; i++) beta.n[i] = GLV_BETA[i];
   gej_set_infinity(R);
   if (progress) *progress = 800;
 #pragma unroll 1
-  for (int w = 63; w >= 0; w--) {
+  for (int w = 32; w >= 0; w--) {
     gej t;
     gej_double(t, R);
     gej_double(R, t);
     gej_double(t, R);
     gej_double(R, t);
-    int word = w >> 4, shift = (w & 15) * 4;
-    u64 dg = (gs.d[word] >> shift) & 15;
-    u64 dp = (ps.d[word] >> shift) & 15;
-    ge gent = KV_G_TABLE[dg];
-    gej_add_ge(t, R, gent);
-    gej_cmov(R, t, (u64)(dg != 0));
-    gej_add(t, R, ptab[dp]);
-    gej_cmov(R, t, (u64)(dp != 0));
+    /* G stream */
+    {
+      u64 d = glv_digit(g1h, w);
+      ge e = KV_G_TABLE[d];
+      fe ny;
+      fe_neg(ny, e.y);
+      fe_cmov(e.y, ny, g1h.neg);
+      gej_add_ge(t, R, e);
+      gej_cmov(R, t, (u64)(d != 0));
+    }
+    /* φG stream: (β·x, ±y) */
+    {
+      u64 d = glv_digit(g2h, w);
+      ge e = KV_G_TABLE[d];
+      fe bx;
+      fe_mul(bx, e.x, beta);
+      e.x = bx;
+      fe ny;
+      fe_neg(ny, e.y);
+      fe_cmov(e.y, ny, g2h.neg);
+      gej_add_ge(t, R, e);
+      gej_cmov(R, t, (u64)(d != 0));
+    }
+    /* P stream */
+    {
+      u64 d = glv_digit(p1h, w);
+      gej e = ptab[d];
+      fe ny;
+      fe_neg(ny, e.y);
+      fe_cmov(e.y, ny, p1h.neg);
+      gej_add(t, R, e);
+      gej_cmov(R, t, (u64)(d != 0));
+    }
+    /* φP stream: (β·X, ±Y, Z) */
+    {
+      u64 d = glv_digit(p2h, w);
+      gej e = ptab[d];
+      fe bx;
+      fe_mul(bx, e.x, beta);
+      e.x = bx;
+      fe ny;
+      fe_neg(ny, e.y);
+      fe_cmov(e.y, ny, p2h.neg);
+      gej_add(t, R, e);
+      gej_cmov(R, t, (u64)(d != 0));
+    }
   }
 }
 
