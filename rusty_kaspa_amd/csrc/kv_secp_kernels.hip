@@ -134,21 +134,24 @@ extern "C" __global__ void kv_ec_table_init_kernel() {
 __device__ __constant__ static const u64 GLV_BETA[4] = {
     0xc1396c28719501eeULL, 0x9cf0497512f58995ULL, 0x6e64479eac3434e9ULL,
     0x7ae96a2b657c0710ULL};
-__device__ __constant__ static const u64 GLV_G1[4] = { /* round(2^384·b2/n) */
+__device__ __constant__ static const u64 GLV_G1[4] = {
     0xe893209a45dbb031ULL, 0x3daa8a1471e8ca7fULL, 0xe86c90e49284eb15ULL,
     0x3086d221a7d46bcdULL};
-__device__ __constant__ static const u64 GLV_G2[4] = { /* round(2^384·(−b1)/n) */
+__device__ __constant__ static const u64 GLV_G2[4] = {
     0x1571b4ae8ac47f71ULL, 0x221208ac9df506c6ULL, 0x6f547fa90abfe4c4ULL,
     0xe4437ed6010e8828ULL};
 __device__ __constant__ static const u64 GLV_A1[4] = {
-    0xe86c90e49284eb15ULL, 0x3086d221a7d46bcdULL, 0, 0};
-__device__ __constant__ static const u64 GLV_B1[4] = { /* −b1 as two's compl. */
+    0xe86c90e49284eb15ULL, 0x3086d221a7d46bcdULL, 0x0000000000000000ULL,
+    0x0000000000000000ULL};
+__device__ __constant__ static const u64 GLV_B1[4] = {
     0x90ab8056f5401b3dULL, 0x1bbc8129fef177d7ULL, 0xffffffffffffffffULL,
     0xffffffffffffffffULL};
 __device__ __constant__ static const u64 GLV_A2[4] = {
-    0x657c1108d9d44cfdULL, 0x14ca50f7a8e2f3f6ULL, 1, 0};
+    0x57c1108d9d44cfd8ULL, 0x14ca50f7a8e2f3f6ULL, 0x0000000000000001ULL,
+    0x0000000000000000ULL};
 __device__ __constant__ static const u64 GLV_B2[4] = {
-    0xe86c90e49284eb15ULL, 0x3086d221a7d46bcdULL, 0, 0};
+    0xe86c90e49284eb15ULL, 0x3086d221a7d46bcdULL, 0x0000000000000000ULL,
+    0x0000000000000000ULL};
 
 /* low-256 product r = (a*b) mod 2^256 (two's-complement arithmetic) */
 __device__ __forceinline__ void mul_low256(u64 r[4], const u64 a[4], const u64 b[4]) {
@@ -187,13 +190,16 @@ __device__ inline void glv_split(const sc &k, glv_half &h1, glv_half &h2) {
 #pragma unroll
     for (int i = 0; i < 4; i++) k1[i] = subb(k1[i], t2[i], borrow);
   }
-  /* k2 = −(c1·b1_neg·(−1) …) — with GLV_B1 = −b1: k2 = c1·(−b1) − c2·b2 */
+  /* k2 = −(c1·b1 + c2·b2), all mod 2^256 two's complement */
   mul_low256(t1, c1, GLV_B1);
   mul_low256(t2, c2, GLV_B2);
   {
+    u64 carry = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++) t1[i] = addc(t1[i], t2[i], carry);
     u64 borrow = 0;
 #pragma unroll
-    for (int i = 0; i < 4; i++) k2[i] = subb(t1[i], t2[i], borrow);
+    for (int i = 0; i < 4; i++) k2[i] = subb(0, t1[i], borrow);
   }
   /* two's-complement abs */
   u64 n1 = k1[3] >> 63, n2 = k2[3] >> 63;
