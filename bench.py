@@ -34,11 +34,11 @@ INVALID_PERMILLE = 0  # all-valid variant is the headline; 10%-invalid via flag
 # Algorithmic work accounting for the roofline (documented in DESIGN.md §4):
 # GLV-split 4-bit windowed ladder: 132 Jacobian doubles (7 fe_mul-equiv) +
 # 66 mixed adds (11) + 66 full adds (16) + 66 φ-multiplies + 15-entry P-table
-# build (14×11) + scalar decomposition + x-lift sqrt (~380) + final inversion
-# (~380) ≈ 3,650 256-bit field multiplies; each fe_mul ≈ 170 u32-ALU-op
-# equivalents (64 32×32 mult-equiv + carries + fold)
-# → ≈ 0.62e6 u32-op-equivalents per verify.
-ALG_OPS_PER_VERIFY = 0.62e6
+# build (14×11) + scalar decomposition + addition-chain x-lift sqrt (266) and
+# final inversion (269) ≈ 3,170 256-bit field multiplies; each fe_mul ≈ 170
+# u32-ALU-op equivalents (64 32×32 mult-equiv + carries + fold)
+# → ≈ 0.54e6 u32-op-equivalents per verify.
+ALG_OPS_PER_VERIFY = 0.54e6
 # gfx950 VALU peak: 256 CU × 4 SIMD × 32 lanes × 2.4 GHz = 78.6 T u32-ops/s
 VALU_PEAK_TOPS = 78.6
 
