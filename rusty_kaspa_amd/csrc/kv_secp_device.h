@@ -452,25 +452,58 @@ __device__ inline void sc_mul(sc &r, const sc &a, const sc &b) {
   sc_reduce8(r, t);
 }
 
-/* scalar inverse mod n via Fermat (ECDSA only) */
+__device__ KV_GROUP_ATTR void sc_sqrn(sc &r, int n) {
+#pragma unroll 1
+  for (int i = 0; i < n; i++) sc_mul(r, r, r);
+}
+
+/* scalar inverse mod n via Fermat (ECDSA only): n-2 = (2^124-1)·2^132 + tail.
+ * All-ones head by addition chain (124 sqr + 10 mul), 132-bit tail by 4-bit
+ * fixed windows of the constant exponent (132 sqr + 33 + 14 table muls)
+ * ≈ 313 sc_mul vs ~512 for the naive 256-step square-and-multiply. The window
+ * digits and the head split are derived and checked symbolically in python
+ * before transcription. */
 __device__ KV_GROUP_ATTR void sc_inv(sc &r, const sc &a) {
-  static const u64 NM2[4] = {0xBFD25E8CD036413FULL, 0xBAAEDCE6AF48A03BULL,
-                             0xFFFFFFFFFFFFFFFEULL, 0xFFFFFFFFFFFFFFFFULL};
-  sc result = {{1, 0, 0, 0}};
-  sc base = a;
+  static const uint8_t WIN[33] = {14, 11, 10, 10, 14, 13, 12, 14, 6,  10, 15,
+                                  4,  8,  10, 0,  3,  11, 11, 15, 13, 2,  5,
+                                  14, 8,  12, 13, 0,  3,  6,  4,  1,  3,  15};
+  sc tab[16];
+  tab[0] = {{1, 0, 0, 0}};
+  tab[1] = a;
 #pragma unroll 1
-  for (int w = 3; w >= 0; w--) {
+  for (int i = 2; i < 16; i++) sc_mul(tab[i], tab[i - 1], a);
+  /* head: x124 = a^(2^124-1); xk here means a^(2^k-1), so x3 = a^7 and
+   * x4 = a^15 — both already in the window table */
+  sc x6, x12, x24, t;
+  const sc &x3 = tab[7], &x4 = tab[15];
+  t = x3;
+  sc_sqrn(t, 3);
+  sc_mul(x6, t, x3);
+  t = x6;
+  sc_sqrn(t, 6);
+  sc_mul(x12, t, x6);
+  t = x12;
+  sc_sqrn(t, 12);
+  sc_mul(x24, t, x12);
+  t = x24;
+  sc_sqrn(t, 24);
+  sc_mul(t, t, x24); /* x48 */
+  sc x48 = t;
+  sc_sqrn(t, 48);
+  sc_mul(t, t, x48); /* x96 */
+  sc_sqrn(t, 24);
+  sc_mul(t, t, x24); /* x120 */
+  sc_sqrn(t, 4);
+  sc_mul(t, t, x4); /* x124 */
+  /* tail: 33 4-bit windows */
 #pragma unroll 1
-    for (int b = 63; b >= 0; b--) {
-      sc_mul(result, result, result);
-      sc t;
-      sc_mul(t, result, base);
-      u64 bit = (NM2[w] >> b) & 1;
-      u64 mask = 0 - bit;
-      for (int k = 0; k < 4; k++) result.d[k] = (result.d[k] & ~mask) | (t.d[k] & mask);
-    }
+  for (int i = 0; i < 33; i++) {
+    sc_sqrn(t, 4);
+    sc m;
+    sc_mul(m, t, tab[WIN[i]]);
+    t = m;
   }
-  r = result;
+  r = t;
 }
 
 /* ---------- group: Jacobian points, a=0 b=7 curve ----------
