@@ -81,6 +81,8 @@ extern "C" {
 #define KV_ERR_SEQUENCE_LOCK 6
 #define KV_ERR_FEERATE_TOO_LOW 7
 #define KV_ERR_MASS_INCOMPUTABLE 8
+#define KV_ERR_MISSING_OUTPOINT 9   /* populate failure: outpoint not in the UTXO set
+                                       (ruleerrors::RuleError::MissingTxOutpoints) */
 #define KV_ERR_BAD_BLOB 90
 #define KV_ERR_SIGNATURE_INVALID_BASE 100
 #define KV_ERR_SIGNATURE_EMPTY_BASE 200
@@ -212,6 +214,20 @@ int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n);
 int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
                    uint8_t *entries_out /*[64n] or NULL*/, uint64_t *found_bitmap,
                    double *kernel_ms /*optional*/);
+
+/* Populate + validate + (optionally) apply the UTXO diff — the full virtual
+ * processor step for one block (utxo_validation.rs:351-390 populate, then
+ * validation, then utxo_diff.rs:224 add_transaction per accepted tx).
+ * Blob format as kv_validate_block, but each input's UtxoEntry fields are
+ * IGNORED (builders write zeros with utxo_spk_len = 0); entries resolve from
+ * the GPU-resident table. A missing outpoint pre-fails that tx with
+ * KV_ERR_MISSING_OUTPOINT (it is skipped by validation and muhash). When
+ * apply_diff != 0, accepted txs' spent outpoints are removed and their created
+ * outputs upserted with block_daa_score (coinbase txs never enter this path). */
+int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
+                           uint64_t pov_daa_score, uint64_t block_daa_score,
+                           uint32_t flags, int apply_diff, int32_t *tx_codes_out,
+                           uint64_t *fees_out, uint8_t *muhash_partial_out);
 
 /* Sig-cache statistics (crypto/txscript/src/caches.rs:57-82 counters). */
 typedef struct {

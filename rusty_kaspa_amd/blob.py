@@ -112,3 +112,51 @@ def finalize_tx_ids(txs, compute_id) -> None:
     blob = build_blob(txs)
     for i, t in enumerate(txs):
         t["tx_id"] = compute_id(blob, i)
+
+
+def strip_utxo_entries(blob: bytes):
+    """Split a populated blob into (unpopulated_blob, seeds).
+
+    The unpopulated blob has every input's UtxoEntry fields zeroed with
+    spk_len = 0 — the shape a node hands to kv_validate_block_utxo, where
+    entries resolve from the GPU-resident UTXO table instead. `seeds` is the
+    list of (outpoint36, entry64) pairs to kv_utxo_upsert beforehand
+    (packed-64B table layout). Mirrors the populate contract of
+    utxo_validation.rs:351-390.
+    """
+    n_txs, = struct.unpack_from("<I", blob, 0)
+    offs = list(struct.unpack_from(f"<{n_txs}I", blob, 4))
+    seeds = []
+    out_txs = []
+    for t in range(n_txs):
+        off = offs[t]
+        end = offs[t + 1] if t + 1 < n_txs else len(blob)
+        p = off
+        n_in, n_out = struct.unpack_from("<HH", blob, p + 2)
+        payload_len, = struct.unpack_from("<I", blob, p + 36)
+        p += 88 + payload_len
+        chunks = [blob[off:p]]
+        for _ in range(n_in):
+            outpoint = blob[p:p + 36]
+            sig_len, = struct.unpack_from("<I", blob, p + 48)
+            chunks.append(blob[p:p + 52 + sig_len])
+            p += 52 + sig_len
+            amount, daa = struct.unpack_from("<QQ", blob, p)
+            is_cb, has_cov = blob[p + 16], blob[p + 17]
+            spkv, = struct.unpack_from("<H", blob, p + 18)
+            spk_len, = struct.unpack_from("<I", blob, p + 20)
+            spk = blob[p + 24:p + 24 + spk_len]
+            assert spk_len <= 36, "inline-table spk limit"
+            entry64 = struct.pack("<QQHHI", amount, daa, is_cb & 1, spkv,
+                                  spk_len) + spk.ljust(36, b"\0") + bytes(4)
+            seeds.append((outpoint, entry64))
+            chunks.append(bytes(24))  # zero entry, spk_len 0
+            p += 24 + spk_len + (32 if has_cov else 0)
+        chunks.append(blob[p:end])  # outputs verbatim
+        out_txs.append(b"".join(chunks))
+    header = [struct.pack("<I", n_txs)]
+    off = 4 + 4 * n_txs
+    for e in out_txs:
+        header.append(struct.pack("<I", off))
+        off += len(e)
+    return b"".join(header + out_txs), seeds

@@ -802,15 +802,13 @@ extern "C" int kv_sighash_batch(kv_ctx *ctx, const uint8_t *blob, size_t blob_le
   return 0;
 }
 
-extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
-                                 uint64_t pov_daa_score, uint64_t block_daa_score,
-                                 uint32_t flags, int32_t *tx_codes_out,
-                                 uint64_t *fees_out, uint8_t *muhash_partial_out) {
-  if (!ctx) {
-    set_error("kv_validate_block: null ctx");
-    return -1;
-  }
-  std::lock_guard<std::mutex> lk(ctx->mu);
+/* caller holds ctx->mu; pre_codes (optional) pre-fails txs (e.g. missing
+ * outpoints from the populate step) so they skip validation and muhash */
+static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
+                               uint64_t pov_daa_score, uint64_t block_daa_score,
+                               uint32_t flags, int32_t *tx_codes_out,
+                               uint64_t *fees_out, uint8_t *muhash_partial_out,
+                               const int32_t *pre_codes) {
   vector<HTx> txs;
   int n_txs = parse_blob_host(blob, blob_len, txs);
   if (n_txs < 0) {
@@ -824,8 +822,11 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
   std::vector<std::vector<InputPlan>> plans(n_txs);
   std::vector<int32_t> codes(n_txs, 0);
   std::vector<uint64_t> fees(n_txs, 0);
+  if (pre_codes)
+    for (int t = 0; t < n_txs; t++) codes[t] = pre_codes[t];
   for (int t = 0; t < n_txs; t++) {
     HTx &tx = txs[t];
+    if (codes[t]) continue;
     if (h_is_coinbase(tx)) { /* coinbase never enters this path (utxo_validation.rs:297) */
       codes[t] = KV_ERR_BAD_BLOB;
       continue;
@@ -1039,6 +1040,20 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
   return 0;
 }
 
+extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
+                                 uint64_t pov_daa_score, uint64_t block_daa_score,
+                                 uint32_t flags, int32_t *tx_codes_out,
+                                 uint64_t *fees_out, uint8_t *muhash_partial_out) {
+  if (!ctx) {
+    set_error("kv_validate_block: null ctx");
+    return -1;
+  }
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  return validate_block_impl(ctx, blob, blob_len, pov_daa_score, block_daa_score,
+                             flags, tx_codes_out, fees_out, muhash_partial_out,
+                             nullptr);
+}
+
 
 /* ---------------- GPU-resident UTXO set (kv_utxo_kernels.hip) ----------------
  * ⇔ utxo_collection.rs:5 HashMap + the populate/diff steps
@@ -1076,9 +1091,8 @@ static int utxo_stage(kv_ctx *ctx, const uint8_t *outpoints, const uint8_t *valu
   return 0;
 }
 
-extern "C" int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints,
+static int utxo_upsert_nolock(kv_ctx *ctx, const uint8_t *outpoints,
                               const uint8_t *entries64, size_t n) {
-  std::lock_guard<std::mutex> lk(ctx->mu);
   if (!ctx->d_utxo) {
     set_error("kv_utxo_upsert: call kv_utxo_reset first");
     return -1;
@@ -1101,8 +1115,13 @@ extern "C" int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints,
   return 0;
 }
 
-extern "C" int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
+extern "C" int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints,
+                              const uint8_t *entries64, size_t n) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  return utxo_upsert_nolock(ctx, outpoints, entries64, n);
+}
+
+static int utxo_remove_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
   if (!ctx->d_utxo) {
     set_error("kv_utxo_remove: call kv_utxo_reset first");
     return -1;
@@ -1117,10 +1136,14 @@ extern "C" int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
   return 0;
 }
 
-extern "C" int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
+extern "C" int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  return utxo_remove_nolock(ctx, outpoints, n);
+}
+
+static int utxo_lookup_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
                               uint8_t *entries_out, uint64_t *found_bitmap,
                               double *kernel_ms) {
-  std::lock_guard<std::mutex> lk(ctx->mu);
   if (!ctx->d_utxo) {
     set_error("kv_utxo_lookup: call kv_utxo_reset first");
     return -1;
@@ -1153,5 +1176,170 @@ extern "C" int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
   }
   (void)hipEventDestroy(t0);
   (void)hipEventDestroy(t1);
+  return 0;
+}
+
+extern "C" int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
+                              uint8_t *entries_out, uint64_t *found_bitmap,
+                              double *kernel_ms) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  return utxo_lookup_nolock(ctx, outpoints, n, entries_out, found_bitmap,
+                            kernel_ms);
+}
+
+/* ---------------- populate + validate + diff-apply ----------------
+ * ⇔ the virtual processor's populate step (utxo_validation.rs:351-390:
+ * each input's UtxoEntry is resolved from the virtual UTXO set; a missing
+ * outpoint fails the tx) followed by validation and utxo_diff application
+ * (utxo_diff.rs:224 add_transaction: remove spent, add created).
+ *
+ * The incoming blob uses the same format but every input's UtxoEntry fields
+ * are ignored (builders write zeros, utxo_spk_len = 0). Entries come from the
+ * GPU-resident table (kv_utxo_reset/upsert); the engine rebuilds a populated
+ * blob internally (one linear pass — the entry fields are the only part that
+ * moves) and runs the standard pipeline on it. */
+extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
+                                      size_t blob_len, uint64_t pov_daa_score,
+                                      uint64_t block_daa_score, uint32_t flags,
+                                      int apply_diff, int32_t *tx_codes_out,
+                                      uint64_t *fees_out,
+                                      uint8_t *muhash_partial_out) {
+  if (!ctx) {
+    set_error("kv_validate_block_utxo: null ctx");
+    return -1;
+  }
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  if (!ctx->d_utxo) {
+    set_error("kv_validate_block_utxo: call kv_utxo_reset first");
+    return -1;
+  }
+  vector<HTx> txs;
+  int n_txs = parse_blob_host(blob, blob_len, txs);
+  if (n_txs < 0) {
+    set_error("kv_validate_block_utxo: malformed blob");
+    return -1;
+  }
+
+  /* populate: one GPU lookup over every input's outpoint */
+  std::vector<uint8_t> ops;
+  size_t n_in_total = 0;
+  for (auto &tx : txs) n_in_total += tx.inputs.size();
+  ops.reserve(n_in_total * 36);
+  for (auto &tx : txs)
+    for (auto &in : tx.inputs) {
+      ops.insert(ops.end(), in.prev_tx_id, in.prev_tx_id + 32);
+      const uint8_t *ix = (const uint8_t *)&in.prev_index;
+      ops.insert(ops.end(), ix, ix + 4);
+    }
+  std::vector<uint8_t> entries(n_in_total * 64);
+  std::vector<uint64_t> found((n_in_total + 63) / 64, 0);
+  if (n_in_total) {
+    int rc = utxo_lookup_nolock(ctx, ops.data(), n_in_total, entries.data(),
+                                found.data(), nullptr);
+    if (rc) return rc;
+  }
+
+  /* rebuild the blob with populated entries; pre-fail txs with missing inputs */
+  std::vector<int32_t> pre_codes(n_txs, 0);
+  std::vector<uint8_t> pop;
+  pop.reserve(blob_len + n_in_total * 40);
+  pop.resize(4 + 4ull * n_txs);
+  memcpy(pop.data(), blob, 4);
+  size_t in_idx = 0;
+  for (int t = 0; t < n_txs; t++) {
+    const HTx &tx = txs[t];
+    uint32_t new_off = (uint32_t)pop.size();
+    memcpy(pop.data() + 4 + 4ull * t, &new_off, 4);
+    uint32_t tx_end = (t + 1 < n_txs) ? txs[t + 1].off : (uint32_t)blob_len;
+    if (tx.inputs.empty()) { /* coinbase or inputless: copy verbatim */
+      pop.insert(pop.end(), blob + tx.off, blob + tx_end);
+      continue;
+    }
+    /* header + payload */
+    pop.insert(pop.end(), blob + tx.off, blob + tx.inputs[0].rec_off);
+    for (size_t i = 0; i < tx.inputs.size(); i++, in_idx++) {
+      const HInput &in = tx.inputs[i];
+      /* outpoint/sequence/commit/sig_script prefix, verbatim */
+      pop.insert(pop.end(), blob + in.rec_off,
+                 blob + in.rec_off + 52 + in.sig_script_len);
+      int hit = (found[in_idx / 64] >> (in_idx % 64)) & 1;
+      if (!hit && !pre_codes[t]) pre_codes[t] = KV_ERR_MISSING_OUTPOINT;
+      /* entry record (64B packed, kv_utxo_kernels.hip layout) → blob fields */
+      const uint8_t *e = entries.data() + in_idx * 64;
+      uint64_t amount = 0, daa = 0;
+      uint16_t eflags = 0, spkv = 0;
+      uint32_t spk_len = 0;
+      if (hit) {
+        memcpy(&amount, e, 8);
+        memcpy(&daa, e + 8, 8);
+        memcpy(&eflags, e + 16, 2);
+        memcpy(&spkv, e + 18, 2);
+        memcpy(&spk_len, e + 20, 4);
+      }
+      uint8_t hdr[24];
+      memcpy(hdr, &amount, 8);
+      memcpy(hdr + 8, &daa, 8);
+      hdr[16] = (uint8_t)(eflags & 1);
+      hdr[17] = 0; /* covenants out of round-1 scope */
+      memcpy(hdr + 18, &spkv, 2);
+      memcpy(hdr + 20, &spk_len, 4);
+      pop.insert(pop.end(), hdr, hdr + 24);
+      if (spk_len) pop.insert(pop.end(), e + 24, e + 24 + spk_len);
+    }
+    /* outputs region runs to the end of the tx */
+    uint32_t outs_start = tx.outputs.empty()
+                              ? tx_end
+                              : tx.output_offs[0];
+    pop.insert(pop.end(), blob + outs_start, blob + tx_end);
+  }
+
+  int rc = validate_block_impl(ctx, pop.data(), pop.size(), pov_daa_score,
+                               block_daa_score, flags, tx_codes_out, fees_out,
+                               muhash_partial_out, pre_codes.data());
+  if (rc || !apply_diff) return rc;
+
+  /* diff apply for accepted txs: remove spent, upsert created */
+  std::vector<uint8_t> del_ops, add_ops, add_ents;
+  in_idx = 0;
+  for (int t = 0; t < n_txs; t++) {
+    const HTx &tx = txs[t];
+    if (tx_codes_out[t] != 0) {
+      in_idx += tx.inputs.size();
+      continue;
+    }
+    for (auto &in : tx.inputs) {
+      del_ops.insert(del_ops.end(), ops.begin() + in_idx * 36,
+                     ops.begin() + (in_idx + 1) * 36);
+      in_idx++;
+    }
+    for (uint32_t i = 0; i < tx.outputs.size(); i++) {
+      const HOutput &o = tx.outputs[i];
+      if (o.spk_len > 36) {
+        set_error("kv_validate_block_utxo: created spk > 36B (round-1 inline "
+                  "table limit)");
+        return -4;
+      }
+      add_ops.insert(add_ops.end(), tx.tx_id, tx.tx_id + 32);
+      const uint8_t *ix = (const uint8_t *)&i;
+      add_ops.insert(add_ops.end(), ix, ix + 4);
+      uint8_t e[64] = {0};
+      memcpy(e, &o.value, 8);
+      memcpy(e + 8, &block_daa_score, 8);
+      /* flags: coinbase txs never enter this path → bit0 = 0 */
+      memcpy(e + 18, &o.spk_version, 2);
+      memcpy(e + 20, &o.spk_len, 4);
+      memcpy(e + 24, o.spk, o.spk_len);
+      add_ents.insert(add_ents.end(), e, e + 64);
+    }
+  }
+  if (!del_ops.empty()) {
+    rc = utxo_remove_nolock(ctx, del_ops.data(), del_ops.size() / 36);
+    if (rc) return rc;
+  }
+  if (!add_ops.empty()) {
+    rc = utxo_upsert_nolock(ctx, add_ops.data(), add_ents.data(),
+                            add_ops.size() / 36);
+    if (rc) return rc;
+  }
   return 0;
 }

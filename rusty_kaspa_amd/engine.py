@@ -114,3 +114,18 @@ class Engine:
             ctypes.c_uint32(flags), codes, fees, partial)
         self._check(rc)
         return list(codes), list(fees), bytes(partial) if want_muhash else None
+
+    def validate_block_utxo(self, blob: bytes, n_txs: int, pov_daa: int,
+                            block_daa: int, flags: int = 2, apply_diff: bool = True,
+                            want_muhash: bool = True):
+        """Populate from the GPU-resident UTXO table, validate, apply diff."""
+        codes = (ctypes.c_int32 * n_txs)()
+        fees = (ctypes.c_uint64 * n_txs)()
+        partial = (ctypes.c_uint8 * 768)() if want_muhash else None
+        rc = self.lib.kv_validate_block_utxo(
+            ctypes.c_void_p(self.ctx), blob, ctypes.c_size_t(len(blob)),
+            ctypes.c_uint64(pov_daa), ctypes.c_uint64(block_daa),
+            ctypes.c_uint32(flags), ctypes.c_int(1 if apply_diff else 0),
+            codes, fees, partial)
+        self._check(rc)
+        return list(codes), list(fees), bytes(partial) if want_muhash else None
