@@ -1014,9 +1014,12 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
         uint32_t n_cur = (uint32_t)cnt;
         uint64_t *cur = src;
         while (n_cur > 1) {
-          /* wave-cooperative mulmod (one value per wave, limbs in registers):
-           * two passes — wide fan-in then a single wave over ≤512 partials */
-          uint32_t stride = n_cur > 512 ? 512 : 1;
+          /* wave-cooperative mulmod (one value per wave, limbs in registers).
+           * Keep every pass parallel: each wave chains ≤64 values, so the
+           * final pass never degenerates into one wave grinding 512 serial
+           * mulmods (measured 2.5 ms at the old stride schedule). */
+          uint32_t stride = n_cur > 64 ? (n_cur + 63) / 64 : 1;
+          if (stride > 1024) stride = 1024;
           uint32_t waves_per_block = 4; /* 256 threads */
           uint32_t blocks = (stride + waves_per_block - 1) / waves_per_block;
           hipLaunchKernelGGL(kv_u3072_reduce_wave_kernel, dim3(blocks), dim3(256),
