@@ -817,19 +817,25 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
   }
   uint64_t sigop_units = ctx->params.mass_per_sig_op * KVH_UNITS_PER_GRAM;
 
-  /* phase 1: host integer checks + classification */
+  /* phase 1: host integer checks + classification, fanned over the host cores
+   * (⇔ the reference's rayon pool) with per-chunk job lists so the GPU job
+   * order after concatenation equals the serial order */
   std::vector<kv::kv_job> sjobs, ejobs;
   std::vector<std::vector<InputPlan>> plans(n_txs);
   std::vector<int32_t> codes(n_txs, 0);
   std::vector<uint64_t> fees(n_txs, 0);
   if (pre_codes)
     for (int t = 0; t < n_txs; t++) codes[t] = pre_codes[t];
-  for (int t = 0; t < n_txs; t++) {
+  unsigned P1T = kvh_threads((uint32_t)n_txs);
+  uint32_t p1_chunk = ((uint32_t)n_txs + P1T - 1) / P1T;
+  std::vector<std::vector<kv::kv_job>> tl_s(P1T), tl_e(P1T);
+  auto phase1_tx = [&](uint32_t t, std::vector<kv::kv_job> &sjobs,
+                       std::vector<kv::kv_job> &ejobs) {
     HTx &tx = txs[t];
-    if (codes[t]) continue;
+    if (codes[t]) return;
     if (h_is_coinbase(tx)) { /* coinbase never enters this path (utxo_validation.rs:297) */
       codes[t] = KV_ERR_BAD_BLOB;
-      continue;
+      return;
     }
     int code = 0;
     for (auto &in : tx.inputs)
@@ -882,11 +888,52 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
         }
     codes[t] = code;
     fees[t] = total_in - total_out;
-    if (code || flags == KV_FLAGS_SKIP_SCRIPT_CHECKS) continue;
+    if (code || flags == KV_FLAGS_SKIP_SCRIPT_CHECKS) return;
     plans[t].reserve(tx.inputs.size());
     for (uint32_t i = 0; i < tx.inputs.size(); i++)
       plans[t].push_back(
           classify_input(tx, tx.inputs[i], sigop_units, sjobs, ejobs, t, i));
+  };
+  {
+    std::vector<std::thread> th;
+    for (unsigned k = 0; k < P1T; k++) {
+      uint32_t lo = k * p1_chunk,
+               hi = std::min((uint32_t)n_txs, lo + p1_chunk);
+      if (lo >= hi) break;
+      auto work = [&, k, lo, hi]() {
+        for (uint32_t t = lo; t < hi; t++) phase1_tx(t, tl_s[k], tl_e[k]);
+      };
+      if (P1T == 1)
+        work();
+      else
+        th.emplace_back(work);
+    }
+    for (auto &x : th) x.join();
+    /* concatenate chunk job lists (serial order preserved) and rebase the
+     * per-plan job indices by each chunk's offset */
+    std::vector<size_t> s_base(P1T, 0), e_base(P1T, 0);
+    size_t sa = 0, ea = 0;
+    for (unsigned k = 0; k < P1T; k++) {
+      s_base[k] = sa;
+      e_base[k] = ea;
+      sa += tl_s[k].size();
+      ea += tl_e[k].size();
+    }
+    sjobs.reserve(sa);
+    ejobs.reserve(ea);
+    for (unsigned k = 0; k < P1T; k++) {
+      sjobs.insert(sjobs.end(), tl_s[k].begin(), tl_s[k].end());
+      ejobs.insert(ejobs.end(), tl_e[k].begin(), tl_e[k].end());
+    }
+    if (P1T > 1)
+      for (int t = 0; t < n_txs; t++) {
+        unsigned k = (uint32_t)t / p1_chunk;
+        for (auto &pl : plans[t]) {
+          if (pl.job >= 0) pl.job += (int32_t)(pl.ecdsa ? e_base[k] : s_base[k]);
+          for (auto &ms : pl.msig_sigs)
+            if (ms.job0 >= 0) ms.job0 += (int32_t)s_base[k];
+        }
+      }
   }
 
   /* phase 2: GPU — subhashes, sighash+tuple assembly, EC verify */
@@ -944,9 +991,10 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
   }
 
-  /* phase 3: resolution (first failing input wins, sequential semantics) */
-  for (int t = 0; t < n_txs; t++) {
-    if (codes[t] || flags == KV_FLAGS_SKIP_SCRIPT_CHECKS) continue;
+  /* phase 3: resolution (first failing input wins, sequential semantics);
+   * read-only over the GPU statuses → fans over the host cores */
+  kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
+    if (codes[t] || flags == KV_FLAGS_SKIP_SCRIPT_CHECKS) return;
     for (uint32_t i = 0; i < txs[t].inputs.size(); i++) {
       int c = resolve_input(plans[t][i], txs[t].inputs[i], sigop_units,
                             s_status.data(), e_status.data());
@@ -955,7 +1003,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
         break;
       }
     }
-  }
+  });
 
   /* phase 4: muhash over valid txs (spends → denominator, creates → numerator) */
   if (muhash_partial_out) {
