@@ -895,20 +895,16 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
           classify_input(tx, tx.inputs[i], sigop_units, sjobs, ejobs, t, i));
   };
   {
-    std::vector<std::thread> th;
-    for (unsigned k = 0; k < P1T; k++) {
-      uint32_t lo = k * p1_chunk,
-               hi = std::min((uint32_t)n_txs, lo + p1_chunk);
-      if (lo >= hi) break;
-      auto work = [&, k, lo, hi]() {
+    if (P1T == 1) {
+      for (uint32_t t = 0; t < (uint32_t)n_txs; t++)
+        phase1_tx(t, tl_s[0], tl_e[0]);
+    } else {
+      KvhPool::inst().run([&](unsigned k) {
+        uint32_t lo = k * p1_chunk,
+                 hi = std::min((uint32_t)n_txs, lo + p1_chunk);
         for (uint32_t t = lo; t < hi; t++) phase1_tx(t, tl_s[k], tl_e[k]);
-      };
-      if (P1T == 1)
-        work();
-      else
-        th.emplace_back(work);
+      });
     }
-    for (auto &x : th) x.join();
     /* concatenate chunk job lists (serial order preserved) and rebase the
      * per-plan job indices by each chunk's offset */
     std::vector<size_t> s_base(P1T, 0), e_base(P1T, 0);
