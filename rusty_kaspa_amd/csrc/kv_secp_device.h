@@ -194,6 +194,19 @@ __device__ __forceinline__ void fe_mul(fe &r, const fe &a, const fe &b) {
   fe_reduce8(r, t);
 }
 
+/* Two INDEPENDENT multiplies in one straight-line body: the backend scheduler
+ * interleaves the two carry chains (separate carry SGPRs), hiding the
+ * VALU-carry hazards that serialize a lone fe_mul. Outputs may alias later
+ * inputs — both products are fully accumulated before either reduce writes. */
+__device__ __forceinline__ void fe_mul2(fe &r1, const fe &a1, const fe &b1,
+                                        fe &r2, const fe &a2, const fe &b2) {
+  u64 t1[8], t2[8];
+  fe_mul_inner(t1, a1.n, b1.n);
+  fe_mul_inner(t2, a2.n, b2.n);
+  fe_reduce8(r1, t1);
+  fe_reduce8(r2, t2);
+}
+
 __device__ __forceinline__ void fe_sqr(fe &r, const fe &a) { fe_mul(r, a, a); }
 
 __device__ __forceinline__ void fe_mul_small(fe &r, const fe &a, u64 k) {
@@ -532,17 +545,16 @@ __device__ __forceinline__ int gej_is_infinity(const gej &a) {
 /* doubling: straight-line, valid for z==0 (result keeps z==0).
  * y == 0 cannot occur on secp256k1 (no 2-torsion). */
 __device__ KV_GROUP_ATTR void gej_double(gej &r, const gej &a) {
-  fe A, B, C, D, E, F, t;
-  fe_sqr(A, a.x);
-  fe_sqr(B, a.y);
-  fe_sqr(C, B);
+  /* 7 muls as 3 interleaved pairs + 1 (fe_mul2 hides the carry-chain stalls) */
+  fe A, B, C, D, E, F, t, zz;
+  fe_mul2(A, a.x, a.x, B, a.y, a.y);
+  fe_mul2(C, B, B, zz, a.y, a.z);
   fe_add(t, a.x, B);
-  fe_sqr(t, t);
+  fe_mul_small(E, A, 3);
+  fe_mul2(t, t, t, F, E, E);
   fe_sub(t, t, A);
   fe_sub(t, t, C);
   fe_add(D, t, t);
-  fe_mul_small(E, A, 3);
-  fe_sqr(F, E);
   fe nx, ny, nz;
   fe_sub(nx, F, D);
   fe_sub(nx, nx, D);
@@ -551,8 +563,7 @@ __device__ KV_GROUP_ATTR void gej_double(gej &r, const gej &a) {
   fe C8;
   fe_mul_small(C8, C, 8);
   fe_sub(ny, t, C8);
-  fe_mul(t, a.y, a.z);
-  fe_add(nz, t, t);
+  fe_add(nz, zz, zz);
   r.x = nx;
   r.y = ny;
   r.z = nz;
@@ -563,11 +574,10 @@ __device__ KV_GROUP_ATTR void gej_double(gej &r, const gej &a) {
  * finite) take a divergent slow path only when a lane actually hits them. */
 __device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
   u64 a_inf = (u64)fe_is_zero(a.z);
+  /* 11 muls as 5 interleaved pairs + 1 */
   fe z1z1, u2, s2, h, hh, i, j, rr, v, t;
-  fe_sqr(z1z1, a.z);
-  fe_mul(u2, b.x, z1z1);
-  fe_mul(s2, b.y, a.z);
-  fe_mul(s2, s2, z1z1);
+  fe_mul2(z1z1, a.z, a.z, s2, b.y, a.z);
+  fe_mul2(u2, b.x, z1z1, s2, s2, z1z1);
   fe_sub(h, u2, a.x);
   fe_sub(rr, s2, a.y);
   if (!a_inf && fe_is_zero(h)) {
@@ -579,26 +589,23 @@ __device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
     }
     return;
   }
-  fe_sqr(hh, h);
+  fe zz;
+  fe_add(zz, a.z, h);
+  fe_mul2(hh, h, h, zz, zz, zz);
   fe_add(i, hh, hh);
   fe_add(i, i, i);
-  fe_mul(j, h, i);
+  fe_mul2(j, h, i, v, a.x, i);
   fe_add(rr, rr, rr);
-  fe_mul(v, a.x, i);
   fe nx, ny, nz;
-  fe_sqr(nx, rr);
+  fe y1j;
+  fe_mul2(nx, rr, rr, y1j, a.y, j);
   fe_sub(nx, nx, j);
   fe_sub(nx, nx, v);
   fe_sub(nx, nx, v);
   fe_sub(t, v, nx);
   fe_mul(t, rr, t);
-  fe y1j;
-  fe_mul(y1j, a.y, j);
   fe_add(y1j, y1j, y1j);
   fe_sub(ny, t, y1j);
-  fe zz;
-  fe_add(zz, a.z, h);
-  fe_sqr(zz, zz);
   fe_sub(zz, zz, z1z1);
   fe_sub(zz, zz, hh);
   nz = zz;
@@ -626,15 +633,12 @@ __device__ KV_GROUP_ATTR void gej_double4(gej &r, const gej &a) {
  * gej_add_ge. b with z==0 produces garbage that callers discard via cmov. */
 __device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
   u64 a_inf = (u64)fe_is_zero(a.z);
+  /* 16 muls as 8 interleaved pairs */
   fe z1z1, z2z2, u1, u2, s1, s2, h, i, j, rr, v, t;
-  fe_sqr(z1z1, a.z);
-  fe_sqr(z2z2, b.z);
-  fe_mul(u1, a.x, z2z2);
-  fe_mul(u2, b.x, z1z1);
-  fe_mul(s1, a.y, b.z);
-  fe_mul(s1, s1, z2z2);
-  fe_mul(s2, b.y, a.z);
-  fe_mul(s2, s2, z1z1);
+  fe_mul2(z1z1, a.z, a.z, z2z2, b.z, b.z);
+  fe_mul2(u1, a.x, z2z2, u2, b.x, z1z1);
+  fe_mul2(s1, a.y, b.z, s2, b.y, a.z);
+  fe_mul2(s1, s1, z2z2, s2, s2, z1z1);
   fe_sub(h, u2, u1);
   fe_sub(rr, s2, s1);
   if (!a_inf && !fe_is_zero(b.z) && fe_is_zero(h)) {
@@ -646,28 +650,24 @@ __device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
     return;
   }
   fe_add(rr, rr, rr); /* r = 2(S2-S1) */
-  fe_sqr(i, h);
+  fe zz;
+  fe_add(zz, a.z, b.z);
+  fe_mul2(i, h, h, zz, zz, zz);
   fe_add(i, i, i);
   fe_add(i, i, i); /* I = 4H^2 */
-  fe_mul(j, h, i);
-  fe_mul(v, u1, i);
+  fe_mul2(j, h, i, v, u1, i);
   fe nx, ny, nz;
-  fe_sqr(nx, rr);
+  fe s1j;
+  fe_mul2(nx, rr, rr, s1j, s1, j);
   fe_sub(nx, nx, j);
   fe_sub(nx, nx, v);
   fe_sub(nx, nx, v);
   fe_sub(t, v, nx);
-  fe_mul(t, rr, t);
-  fe s1j;
-  fe_mul(s1j, s1, j);
-  fe_add(s1j, s1j, s1j);
-  fe_sub(ny, t, s1j);
-  fe zz;
-  fe_add(zz, a.z, b.z);
-  fe_sqr(zz, zz);
   fe_sub(zz, zz, z1z1);
   fe_sub(zz, zz, z2z2);
-  fe_mul(nz, zz, h);
+  fe_mul2(t, rr, t, nz, zz, h);
+  fe_add(s1j, s1j, s1j);
+  fe_sub(ny, t, s1j);
   /* a infinity → result = b */
   fe_cmov(nx, b.x, a_inf);
   fe_cmov(ny, b.y, a_inf);
