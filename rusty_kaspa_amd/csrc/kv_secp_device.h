@@ -25,11 +25,19 @@ namespace kv {
 
 /* Build-time tuning knobs (perf experiments; default = the proven-safe config):
  *  -DKV_GROUP_INLINE   : inline the group ops instead of __noinline__ calls
- *                        (removes ABI spills; loops stay rolled via unroll 1) */
+ *                        (removes ABI spills; loops stay rolled via unroll 1)
+ *  -DKV_DOUBLE_INLINE  : inline ONLY gej_double (half the ladder's calls)
+ * Full inlining reproducibly hangs gfx950 (long-branch relaxation on the
+ * giant body); gej_double alone is the bounded experiment. */
 #ifdef KV_GROUP_INLINE
 #define KV_GROUP_ATTR inline
 #else
 #define KV_GROUP_ATTR __noinline__
+#endif
+#if defined(KV_GROUP_INLINE) || defined(KV_DOUBLE_INLINE)
+#define KV_DOUBLE_ATTR inline
+#else
+#define KV_DOUBLE_ATTR __noinline__
 #endif
 
 typedef uint64_t u64;
@@ -207,7 +215,94 @@ __device__ __forceinline__ void fe_mul2(fe &r1, const fe &a1, const fe &b1,
   fe_reduce8(r2, t2);
 }
 
-__device__ __forceinline__ void fe_sqr(fe &r, const fe &a) { fe_mul(r, a, a); }
+/* dedicated squaring: 6 cross + 4 diagonal 64×64 products (vs 16 for the
+ * generic multiply) — ~48% of the ladder's field muls are squarings. Cross
+ * products accumulate un-doubled (every step adds ≤ product + 2 u64, safely
+ * inside u128), then one shift-chain doubles them before the diagonals. */
+__device__ __forceinline__ void fe_sqr_inner(u64 t[8], const u64 *a) {
+  u64 t1, t2, t3, t4, t5, t6, c;
+  u128 m;
+  m = (u128)a[0] * a[1];
+  t1 = (u64)m;
+  c = (u64)(m >> 64);
+  m = (u128)a[0] * a[2] + c;
+  t2 = (u64)m;
+  c = (u64)(m >> 64);
+  m = (u128)a[0] * a[3] + c;
+  t3 = (u64)m;
+  t4 = (u64)(m >> 64);
+  m = (u128)a[1] * a[2] + t3;
+  t3 = (u64)m;
+  c = (u64)(m >> 64);
+  m = (u128)a[1] * a[3] + t4 + c;
+  t4 = (u64)m;
+  t5 = (u64)(m >> 64);
+  m = (u128)a[2] * a[3] + t5;
+  t5 = (u64)m;
+  t6 = (u64)(m >> 64);
+  /* double the cross part */
+  u64 t7 = t6 >> 63;
+  t6 = (t6 << 1) | (t5 >> 63);
+  t5 = (t5 << 1) | (t4 >> 63);
+  t4 = (t4 << 1) | (t3 >> 63);
+  t3 = (t3 << 1) | (t2 >> 63);
+  t2 = (t2 << 1) | (t1 >> 63);
+  t1 <<= 1;
+  /* diagonals into even columns */
+  u64 lo0, hi0, lo1, hi1, lo2, hi2, lo3, hi3;
+  m = (u128)a[0] * a[0];
+  lo0 = (u64)m;
+  hi0 = (u64)(m >> 64);
+  m = (u128)a[1] * a[1];
+  lo1 = (u64)m;
+  hi1 = (u64)(m >> 64);
+  m = (u128)a[2] * a[2];
+  lo2 = (u64)m;
+  hi2 = (u64)(m >> 64);
+  m = (u128)a[3] * a[3];
+  lo3 = (u64)m;
+  hi3 = (u64)(m >> 64);
+  t[0] = lo0;
+  m = (u128)t1 + hi0;
+  t[1] = (u64)m;
+  m = (m >> 64) + t2 + lo1;
+  t[2] = (u64)m;
+  m = (m >> 64) + t3 + hi1;
+  t[3] = (u64)m;
+  m = (m >> 64) + t4 + lo2;
+  t[4] = (u64)m;
+  m = (m >> 64) + t5 + hi2;
+  t[5] = (u64)m;
+  m = (m >> 64) + t6 + lo3;
+  t[6] = (u64)m;
+  m = (m >> 64) + t7 + hi3;
+  t[7] = (u64)m;
+}
+
+__device__ __forceinline__ void fe_sqr(fe &r, const fe &a) {
+  u64 t[8];
+  fe_sqr_inner(t, a.n);
+  fe_reduce8(r, t);
+}
+
+/* paired variants (same interleaving rationale as fe_mul2) */
+__device__ __forceinline__ void fe_sqr2(fe &r1, const fe &a1, fe &r2,
+                                        const fe &a2) {
+  u64 t1[8], t2[8];
+  fe_sqr_inner(t1, a1.n);
+  fe_sqr_inner(t2, a2.n);
+  fe_reduce8(r1, t1);
+  fe_reduce8(r2, t2);
+}
+
+__device__ __forceinline__ void fe_sqr_mul(fe &rs, const fe &as, fe &rm,
+                                           const fe &am, const fe &bm) {
+  u64 t1[8], t2[8];
+  fe_sqr_inner(t1, as.n);
+  fe_mul_inner(t2, am.n, bm.n);
+  fe_reduce8(rs, t1);
+  fe_reduce8(rm, t2);
+}
 
 __device__ __forceinline__ void fe_mul_small(fe &r, const fe &a, u64 k) {
   u128 c = (u128)a.n[0] * k;
@@ -544,14 +639,14 @@ __device__ __forceinline__ int gej_is_infinity(const gej &a) {
 
 /* doubling: straight-line, valid for z==0 (result keeps z==0).
  * y == 0 cannot occur on secp256k1 (no 2-torsion). */
-__device__ KV_GROUP_ATTR void gej_double(gej &r, const gej &a) {
+__device__ __forceinline__ void gej_double_impl(gej &r, const gej &a) {
   /* 7 muls as 3 interleaved pairs + 1 (fe_mul2 hides the carry-chain stalls) */
   fe A, B, C, D, E, F, t, zz;
-  fe_mul2(A, a.x, a.x, B, a.y, a.y);
-  fe_mul2(C, B, B, zz, a.y, a.z);
+  fe_sqr2(A, a.x, B, a.y);
+  fe_sqr_mul(C, B, zz, a.y, a.z);
   fe_add(t, a.x, B);
   fe_mul_small(E, A, 3);
-  fe_mul2(t, t, t, F, E, E);
+  fe_sqr2(t, t, F, E);
   fe_sub(t, t, A);
   fe_sub(t, t, C);
   fe_add(D, t, t);
@@ -569,6 +664,10 @@ __device__ KV_GROUP_ATTR void gej_double(gej &r, const gej &a) {
   r.z = nz;
 }
 
+__device__ KV_DOUBLE_ATTR void gej_double(gej &r, const gej &a) {
+  gej_double_impl(r, a);
+}
+
 /* mixed add r = a + B(affine): generic madd formulas computed unconditionally;
  * a==infinity fixed up with selects; the rare equal-x cases (h==0 with a
  * finite) take a divergent slow path only when a lane actually hits them. */
@@ -576,7 +675,7 @@ __device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
   u64 a_inf = (u64)fe_is_zero(a.z);
   /* 11 muls as 5 interleaved pairs + 1 */
   fe z1z1, u2, s2, h, hh, i, j, rr, v, t;
-  fe_mul2(z1z1, a.z, a.z, s2, b.y, a.z);
+  fe_sqr_mul(z1z1, a.z, s2, b.y, a.z);
   fe_mul2(u2, b.x, z1z1, s2, s2, z1z1);
   fe_sub(h, u2, a.x);
   fe_sub(rr, s2, a.y);
@@ -591,14 +690,14 @@ __device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
   }
   fe zz;
   fe_add(zz, a.z, h);
-  fe_mul2(hh, h, h, zz, zz, zz);
+  fe_sqr2(hh, h, zz, zz);
   fe_add(i, hh, hh);
   fe_add(i, i, i);
   fe_mul2(j, h, i, v, a.x, i);
   fe_add(rr, rr, rr);
   fe nx, ny, nz;
   fe y1j;
-  fe_mul2(nx, rr, rr, y1j, a.y, j);
+  fe_sqr_mul(nx, rr, y1j, a.y, j);
   fe_sub(nx, nx, j);
   fe_sub(nx, nx, v);
   fe_sub(nx, nx, v);
@@ -621,12 +720,15 @@ __device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
 
 
 /* four doublings in one call — quarters the ABI crossings of the window ladder */
+/* Four doublings in ONE call frame: the accumulator crosses the noinline ABI
+ * (scratch round-trip) once per window instead of four times. The impl is
+ * straight-line, so SROA keeps the intermediate points in VGPRs. */
 __device__ KV_GROUP_ATTR void gej_double4(gej &r, const gej &a) {
   gej t;
-  gej_double(t, a);
-  gej_double(r, t);
-  gej_double(t, r);
-  gej_double(r, t);
+  gej_double_impl(t, a);
+  gej_double_impl(r, t);
+  gej_double_impl(t, r);
+  gej_double_impl(r, t);
 }
 
 /* full Jacobian + Jacobian add (add-2007-bl shape), same select discipline as
@@ -635,7 +737,7 @@ __device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
   u64 a_inf = (u64)fe_is_zero(a.z);
   /* 16 muls as 8 interleaved pairs */
   fe z1z1, z2z2, u1, u2, s1, s2, h, i, j, rr, v, t;
-  fe_mul2(z1z1, a.z, a.z, z2z2, b.z, b.z);
+  fe_sqr2(z1z1, a.z, z2z2, b.z);
   fe_mul2(u1, a.x, z2z2, u2, b.x, z1z1);
   fe_mul2(s1, a.y, b.z, s2, b.y, a.z);
   fe_mul2(s1, s1, z2z2, s2, s2, z1z1);
@@ -652,13 +754,13 @@ __device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
   fe_add(rr, rr, rr); /* r = 2(S2-S1) */
   fe zz;
   fe_add(zz, a.z, b.z);
-  fe_mul2(i, h, h, zz, zz, zz);
+  fe_sqr2(i, h, zz, zz);
   fe_add(i, i, i);
   fe_add(i, i, i); /* I = 4H^2 */
   fe_mul2(j, h, i, v, u1, i);
   fe nx, ny, nz;
   fe s1j;
-  fe_mul2(nx, rr, rr, s1j, s1, j);
+  fe_sqr_mul(nx, rr, s1j, s1, j);
   fe_sub(nx, nx, j);
   fe_sub(nx, nx, v);
   fe_sub(nx, nx, v);

@@ -263,10 +263,7 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const g
 #pragma unroll 1
   for (int w = 32; w >= 0; w--) {
     gej t;
-    gej_double(t, R);
-    gej_double(R, t);
-    gej_double(t, R);
-    gej_double(R, t);
+    gej_double4(R, R);
     /* G stream */
     {
       u64 d = glv_digit(g1h, w);
