@@ -330,6 +330,256 @@ __device__ __forceinline__ void fe_mul_small(fe &r, const fe &a, u64 k) {
   fe_norm_once(r);
 }
 
+
+/* ================= 10x26 field representation (fe26) =================
+ * (limb convention: l0..l8 < m*2^26, l9 < m*2^22 for magnitude m)
+ *
+ * The 4xu64 representation above lowers every limb product into
+ * v_mad_u64_u32 + v_add_co/v_addc VCC-carry chains; measured disassembly of
+ * the group ops shows 24% s_nop (VCC read-after-write padding) + 30% carry
+ * adds + 16% movs — only ~10% of issue slots multiply. The 10x26 form
+ * (limbs < 2^26, base 2^26, 19 column sums < 2^62 for operand magnitude <= 8)
+ * accumulates whole columns in plain 64-bit v_mad_u64_u32 chains with NO
+ * carry flags at all, eliminating the hazard padding. Derived and fuzzed
+ * against exact bigint arithmetic (20k cases, every intermediate < 2^64)
+ * before transcription.
+ *
+ * Magnitude discipline (limb_i < m * 2^26, top < m * 2^26):
+ *   - fe26_mul / fe26_sqr outputs: m = 1 (top limb may graze 2^26: m <= 2)
+ *   - inputs to mul/sqr MUST have m <= 8
+ *   - fe26_add: m = ma + mb;  fe26_neg(m): result m+1;  adds/subs may run to
+ *     m <= 31 (u32 limb bound) before a fe26_norm_weak (-> m = 1)
+ * Group ops below annotate magnitudes; the host-test build asserts them. */
+
+#define KV26_M ((u32)0x3FFFFFF)
+#define KV26_R0 ((u64)0x3D10) /* 2^260 mod p = R0 + 2^36 (C<<4) */
+
+struct fe26 {
+  u32 l[10];
+};
+
+#ifdef KV_HOST_TEST
+#include <assert.h>
+#define KV26_CHECK_MUL_IN(a)                                                   \
+  do {                                                                         \
+    for (int _i = 0; _i < 10; _i++) assert((a).l[_i] <= 8u * (1u << 26));      \
+  } while (0)
+#else
+#define KV26_CHECK_MUL_IN(a)
+#endif
+
+__device__ __forceinline__ void fe26_reduce19(fe26 &r, const u64 t[19]) {
+  /* fold columns 10..18 into 0..12 via 2^260 == R0 + 2^36 (mod p)·2^(26k) */
+  u64 s[13];
+#pragma unroll
+  for (int k = 0; k < 10; k++) s[k] = t[k];
+  s[10] = s[11] = s[12] = 0;
+#pragma unroll
+  for (int k = 0; k < 9; k++) {
+    u64 lo = t[k + 10] & KV26_M;
+    u64 hi = t[k + 10] >> 26;
+    s[k] += lo * KV26_R0;
+    s[k + 1] += (lo << 10) + hi * KV26_R0;
+    s[k + 2] += hi << 10;
+  }
+  u64 out[13], c = 0;
+#pragma unroll
+  for (int k = 0; k < 13; k++) {
+    u64 d = s[k] + c;
+    out[k] = d & KV26_M;
+    c = d >> 26;
+  }
+  /* second fold: tiny spill in out[10..12] */
+  u64 extra = out[10] | (out[11] << 26) | (out[12] << 52);
+  u64 lo = extra & KV26_M, hi = extra >> 26;
+  out[0] += lo * KV26_R0;
+  out[1] += (lo << 10) + hi * KV26_R0;
+  out[2] += hi << 10;
+  c = 0;
+#pragma unroll
+  for (int k = 0; k < 10; k++) {
+    u64 d = out[k] + c;
+    r.l[k] = (u32)(d & KV26_M);
+    c = d >> 26;
+  }
+  /* residual carry (tiny): 2^260 fold once more */
+  r.l[0] += (u32)(c * KV26_R0);
+  r.l[1] += (u32)(c << 10);
+  /* keep the top limb 22-bit (the magnitude convention scales l9 by 2^22):
+   * x*2^256 == x*(2^32 + 977) */
+  u32 x = r.l[9] >> 22;
+  r.l[9] &= 0x3FFFFF;
+  r.l[0] += x * 977u;
+  r.l[1] += x << 6;
+}
+
+__device__ __forceinline__ void fe26_mul_inner(u64 t[19], const fe26 &a,
+                                               const fe26 &b) {
+  KV26_CHECK_MUL_IN(a);
+  KV26_CHECK_MUL_IN(b);
+#pragma unroll
+  for (int k = 0; k < 19; k++) t[k] = 0;
+#pragma unroll
+  for (int i = 0; i < 10; i++)
+#pragma unroll
+    for (int j = 0; j < 10; j++) t[i + j] += (u64)a.l[i] * b.l[j];
+}
+
+__device__ __forceinline__ void fe26_sqr_inner(u64 t[19], const fe26 &a) {
+  KV26_CHECK_MUL_IN(a);
+#pragma unroll
+  for (int k = 0; k < 19; k++) t[k] = 0;
+#pragma unroll
+  for (int i = 0; i < 10; i++) {
+#pragma unroll
+    for (int j = i + 1; j < 10; j++) t[i + j] += (u64)a.l[i] * a.l[j];
+  }
+#pragma unroll
+  for (int k = 0; k < 19; k++) t[k] <<= 1;
+#pragma unroll
+  for (int i = 0; i < 10; i++) t[2 * i] += (u64)a.l[i] * a.l[i];
+}
+
+__device__ __forceinline__ void fe26_mul(fe26 &r, const fe26 &a, const fe26 &b) {
+  u64 t[19];
+  fe26_mul_inner(t, a, b);
+  fe26_reduce19(r, t);
+}
+
+__device__ __forceinline__ void fe26_sqr(fe26 &r, const fe26 &a) {
+  u64 t[19];
+  fe26_sqr_inner(t, a);
+  fe26_reduce19(r, t);
+}
+
+__device__ __forceinline__ void fe26_add(fe26 &r, const fe26 &a, const fe26 &b) {
+#pragma unroll
+  for (int i = 0; i < 10; i++) r.l[i] = a.l[i] + b.l[i];
+}
+
+/* r = -a for input magnitude <= m (result magnitude m+1) */
+__device__ __forceinline__ void fe26_neg(fe26 &r, const fe26 &a, u32 m) {
+  r.l[0] = (u32)(0x3FFFC2FUL * 2 * (m + 1)) - a.l[0];
+  r.l[1] = (u32)(0x3FFFFBFUL * 2 * (m + 1)) - a.l[1];
+#pragma unroll
+  for (int i = 2; i < 9; i++) r.l[i] = (u32)(0x3FFFFFFUL * 2 * (m + 1)) - a.l[i];
+  r.l[9] = (u32)(0x3FFFFFUL * 2 * (m + 1)) - a.l[9];
+}
+
+__device__ __forceinline__ void fe26_mul_int(fe26 &r, u32 k) {
+#pragma unroll
+  for (int i = 0; i < 10; i++) r.l[i] *= k;
+}
+
+__device__ __forceinline__ void fe26_cmov(fe26 &r, const fe26 &a, u32 cond) {
+  u32 mask = 0 - cond;
+#pragma unroll
+  for (int i = 0; i < 10; i++) r.l[i] = (r.l[i] & ~mask) | (a.l[i] & mask);
+}
+
+/* one carry sweep + 2^260 fold: magnitude -> 1 (l0/l1 may graze 2^26+2^20,
+ * i.e. "magnitude 2" for safety accounting; value stays < 2^260) */
+__device__ __forceinline__ void fe26_norm_weak(fe26 &a) {
+  u32 c = 0;
+#pragma unroll
+  for (int i = 0; i < 10; i++) {
+    u32 d = a.l[i] + c;
+    a.l[i] = d & KV26_M;
+    c = d >> 26;
+  }
+  /* c = value >> 260 (<= input magnitude); 2^260 == R0 + 2^36 (mod p) */
+  a.l[0] += c * (u32)KV26_R0;
+  a.l[1] += c << 10;
+  u32 x = a.l[9] >> 22;
+  a.l[9] &= 0x3FFFFF;
+  a.l[0] += x * 977u;
+  a.l[1] += x << 6;
+}
+
+/* full (canonical) normalize — used on the rare paths only (equality, zero
+ * tests, serialization). Deterministic bounded sequence: sweeps + top folds
+ * until value < 2^256, then a branchless conditional subtract of p via the
+ * "value + C overflows bit 256" test. */
+__device__ __forceinline__ void fe26_normalize(fe26 &a) {
+  fe26_norm_weak(a);
+  /* 3 sweep+fold passes: if a pass folds x>0 the value drops below 2^37-ish,
+   * so the next pass has x=0 and leaves canonical limbs */
+#pragma unroll 1
+  for (int pass = 0; pass < 3; pass++) {
+    u32 c = 0;
+#pragma unroll
+    for (int i = 0; i < 10; i++) {
+      u32 d = a.l[i] + c;
+      a.l[i] = d & KV26_M;
+      c = d >> 26;
+    }
+    /* fold 2^256 overflow of the top limb: x*2^256 == x*(2^32 + 977) */
+    u32 x = (a.l[9] >> 22) | (c << 4);
+    a.l[9] &= 0x3FFFFF;
+    a.l[0] += x * 977u;
+    a.l[1] += x << 6;
+  }
+  /* value < 2^256 now (limbs canonical after one more sweep inside the +C
+   * pass below). value >= p  <=>  value + C >= 2^256 */
+  u32 t[10];
+  u64 d = (u64)a.l[0] + 977u;
+  t[0] = (u32)(d & KV26_M);
+  d = (d >> 26) + a.l[1] + (1u << 6);
+  t[1] = (u32)(d & KV26_M);
+#pragma unroll
+  for (int i = 2; i < 10; i++) {
+    d = (d >> 26) + a.l[i];
+    t[i] = (u32)(d & KV26_M);
+  }
+  u32 ge = (u32)(t[9] >> 22) & 1; /* bit 256 set -> value >= p */
+  t[9] &= 0x3FFFFF;
+  u32 mask = 0 - ge;
+#pragma unroll
+  for (int i = 0; i < 10; i++) a.l[i] = (a.l[i] & ~mask) | (t[i] & mask);
+}
+
+/* does a weakly-reduced value equal 0 mod p? (normalize a copy fully) */
+__device__ __forceinline__ int fe26_is_zero(const fe26 &a) {
+  fe26 t = a;
+  fe26_normalize(t);
+  u32 z = 0;
+#pragma unroll
+  for (int i = 0; i < 10; i++) z |= t.l[i];
+  return z == 0;
+}
+
+__device__ __forceinline__ int fe26_eq(const fe26 &a, const fe26 &b) {
+  fe26 ta = a, tb = b;
+  fe26_normalize(ta);
+  fe26_normalize(tb);
+  u32 d = 0;
+#pragma unroll
+  for (int i = 0; i < 10; i++) d |= ta.l[i] ^ tb.l[i];
+  return d == 0;
+}
+
+/* conversions at the byte/u64 boundary */
+__device__ __forceinline__ void fe26_from_fe(fe26 &r, const fe &a) {
+#pragma unroll
+  for (int i = 0; i < 10; i++) {
+    int bit = 26 * i;
+    int w = bit >> 6, sh = bit & 63;
+    u64 v = a.n[w] >> sh;
+    if (sh > 38 && w < 3) v |= a.n[w + 1] << (64 - sh);
+    r.l[i] = (u32)(v & KV26_M);
+  }
+  r.l[9] &= 0x3FFFFF;
+}
+
+/* input must be fully normalized */
+__device__ __forceinline__ void fe26_to_fe(fe &r, const fe26 &a) {
+  r.n[0] = (u64)a.l[0] | ((u64)a.l[1] << 26) | ((u64)a.l[2] << 52);
+  r.n[1] = ((u64)a.l[2] >> 12) | ((u64)a.l[3] << 14) | ((u64)a.l[4] << 40);
+  r.n[2] = ((u64)a.l[4] >> 24) | ((u64)a.l[5] << 2) | ((u64)a.l[6] << 28) |
+           ((u64)a.l[7] << 54);
+  r.n[3] = ((u64)a.l[7] >> 10) | ((u64)a.l[8] << 16) | ((u64)a.l[9] << 42);
+}
+
 __device__ __forceinline__ int fe_is_zero(const fe &a) {
   return (a.n[0] | a.n[1] | a.n[2] | a.n[3]) == 0;
 }
