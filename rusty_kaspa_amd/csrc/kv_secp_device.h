@@ -358,11 +358,14 @@ struct fe26 {
   u32 l[10];
 };
 
+/* mul/sqr safety bound: limbs <= 2*8*2^26 = 2^30 ("magnitude 8" with the
+ * negate convention limb(m) <= 2m*2^26): 10 column products of (2^30)^2 sum
+ * to < 1.4*2^63 < 2^64. The host-test build asserts it on every call. */
 #ifdef KV_HOST_TEST
 #include <assert.h>
 #define KV26_CHECK_MUL_IN(a)                                                   \
   do {                                                                         \
-    for (int _i = 0; _i < 10; _i++) assert((a).l[_i] <= 8u * (1u << 26));      \
+    for (int _i = 0; _i < 10; _i++) assert((a).l[_i] <= (1u << 30));           \
   } while (0)
 #else
 #define KV26_CHECK_MUL_IN(a)
@@ -864,51 +867,145 @@ __device__ KV_GROUP_ATTR void sc_inv(sc &r, const sc &a) {
   r = t;
 }
 
-/* ---------- group: Jacobian points, a=0 b=7 curve ----------
- * Infinity is represented as Z == 0 (no flag field, no early returns): point
- * structs stay fully SROA-decomposed into VGPRs — address-taken structs with
- * control-flow copies were observed to fall back to flat memory on gfx950. */
+/* ---------- fe26 chains: Fermat powers on the 10x26 form ---------- */
+
+__device__ KV_GROUP_ATTR void fe26_sqrn(fe26 &r, int n) {
+#pragma unroll 1
+  for (int i = 0; i < n; i++) fe26_sqr(r, r);
+}
+
+/* addition-chain blocks a^(2^k-1); same chain as the 4xu64 version above
+ * (verified symbolically), restated on fe26 */
+__device__ KV_GROUP_ATTR void fe26_chain223(fe26 &x223, fe26 &x22, fe26 &x2,
+                                            fe26 &x3, const fe26 &a) {
+  fe26 t, x11, x44, x88;
+  fe26_sqr(t, a);
+  fe26_mul(x2, t, a);
+  fe26_sqr(t, x2);
+  fe26_mul(x3, t, a);
+  t = x3;
+  fe26_sqrn(t, 3);
+  fe26_mul(t, t, x3);
+  fe26_sqrn(t, 3);
+  fe26_mul(t, t, x3);
+  fe26_sqrn(t, 2);
+  fe26_mul(x11, t, x2);
+  t = x11;
+  fe26_sqrn(t, 11);
+  fe26_mul(x22, t, x11);
+  t = x22;
+  fe26_sqrn(t, 22);
+  fe26_mul(x44, t, x22);
+  t = x44;
+  fe26_sqrn(t, 44);
+  fe26_mul(x88, t, x44);
+  t = x88;
+  fe26_sqrn(t, 88);
+  fe26_mul(t, t, x88);
+  fe26_sqrn(t, 44);
+  fe26_mul(t, t, x44);
+  fe26_sqrn(t, 3);
+  fe26_mul(x223, t, x3);
+}
+
+__device__ inline void fe26_inv(fe26 &r, const fe26 &a) {
+  fe26 x223, x22, x2, x3, t;
+  fe26_chain223(x223, x22, x2, x3, a);
+  t = x223;
+  fe26_sqrn(t, 23);
+  fe26_mul(t, t, x22);
+  fe26_sqrn(t, 5);
+  fe26_mul(t, t, a);
+  fe26_sqrn(t, 3);
+  fe26_mul(t, t, x2);
+  fe26_sqrn(t, 2);
+  fe26_mul(r, t, a);
+}
+
+/* sqrt via a^((p+1)/4); returns 1 if r*r == a */
+__device__ inline int fe26_sqrt(fe26 &r, const fe26 &a) {
+  fe26 x223, x22, x2, x3, t, chk;
+  fe26_chain223(x223, x22, x2, x3, a);
+  t = x223;
+  fe26_sqrn(t, 23);
+  fe26_mul(t, t, x22);
+  fe26_sqrn(t, 6);
+  fe26_mul(t, t, x2);
+  fe26_sqrn(t, 2);
+  fe26_sqr(chk, t);
+  r = t;
+  fe26 aa = a;
+  return fe26_eq(chk, aa);
+}
+
+/* ---------- group: Jacobian points, a=0 b=7 curve, fe26 coordinates ----------
+ * Infinity is represented as Z == 0 (no flag field, no early returns).
+ * Magnitude discipline: every group-op OUTPUT coordinate has magnitude <= 2
+ * (fe26_norm_weak where sums exceed it); every mul/sqr INPUT stays <= 8 —
+ * the per-line annotations below track the worst case. */
 
 struct ge {
-  fe x, y; /* affine */
+  fe26 x, y; /* affine */
 };
 
 struct gej {
-  fe x, y, z; /* z == 0 ⇔ infinity */
+  fe26 x, y, z; /* z == 0 ⇔ infinity */
 };
 
+__device__ __forceinline__ void fe26_set_int(fe26 &r, u32 v) {
+  r.l[0] = v;
+#pragma unroll
+  for (int i = 1; i < 10; i++) r.l[i] = 0;
+}
+
 __device__ __forceinline__ void gej_set_infinity(gej &r) {
-  r.x = {{1, 0, 0, 0}};
-  r.y = {{1, 0, 0, 0}};
-  r.z = {{0, 0, 0, 0}};
+  fe26_set_int(r.x, 1);
+  fe26_set_int(r.y, 1);
+  fe26_set_int(r.z, 0);
 }
 
 __device__ __forceinline__ int gej_is_infinity(const gej &a) {
-  return fe_is_zero(a.z);
+  return fe26_is_zero(a.z);
 }
 
 /* doubling: straight-line, valid for z==0 (result keeps z==0).
- * y == 0 cannot occur on secp256k1 (no 2-torsion). */
+ * y == 0 cannot occur on secp256k1 (no 2-torsion). Inputs magnitude <= 2. */
 __device__ __forceinline__ void gej_double_impl(gej &r, const gej &a) {
-  /* 7 muls as 3 interleaved pairs + 1 (fe_mul2 hides the carry-chain stalls) */
-  fe A, B, C, D, E, F, t, zz;
-  fe_sqr2(A, a.x, B, a.y);
-  fe_sqr_mul(C, B, zz, a.y, a.z);
-  fe_add(t, a.x, B);
-  fe_mul_small(E, A, 3);
-  fe_sqr2(t, t, F, E);
-  fe_sub(t, t, A);
-  fe_sub(t, t, C);
-  fe_add(D, t, t);
-  fe nx, ny, nz;
-  fe_sub(nx, F, D);
-  fe_sub(nx, nx, D);
-  fe_sub(t, D, nx);
-  fe_mul(t, E, t);
-  fe C8;
-  fe_mul_small(C8, C, 8);
-  fe_sub(ny, t, C8);
-  fe_add(nz, zz, zz);
+  fe26 A, B, C, D, E, F, t, zz;
+  fe26_sqr(A, a.x);        /* 1 */
+  fe26_sqr(B, a.y);        /* 1 */
+  fe26_sqr(C, B);          /* 1 */
+  fe26_mul(zz, a.y, a.z);  /* 1 */
+  fe26_add(t, a.x, B);     /* 2+2=4 */
+  fe26_sqr(t, t);          /* 1 */
+  E = A;
+  fe26_mul_int(E, 3);      /* 6 */
+  fe26_sqr(F, E);          /* 1 */
+  fe26 nA, nC;
+  fe26_neg(nA, A, 2);      /* 3 */
+  fe26_neg(nC, C, 2);      /* 3 */
+  fe26_add(t, t, nA);
+  fe26_add(D, t, nC);      /* 1+3+3 = 8 (pre-double) */
+  fe26_add(D, D, D);       /* 16 */
+  fe26_norm_weak(D);       /* -> 2 */
+  fe26 nx, ny, nz, nD;
+  fe26_neg(nD, D, 2);      /* 3 */
+  fe26_add(nx, F, nD);
+  fe26_add(nx, nx, nD);    /* 2+3+3 = 8 */
+  fe26_norm_weak(nx);      /* -> 2 */
+  fe26 nnx;
+  fe26_neg(nnx, nx, 2);    /* 3 */
+  fe26_add(t, D, nnx);     /* 2+3 = 5: mul input ok */
+  fe26_mul(t, E, t);       /* E 6, t 5 -> 1 */
+  fe26 C8;
+  C8 = C;
+  fe26_mul_int(C8, 8);     /* 16 */
+  fe26 nC8;
+  fe26_neg(nC8, C8, 16);   /* 17 */
+  fe26_add(ny, t, nC8);    /* 19 <= 31 */
+  fe26_norm_weak(ny);      /* -> 2 */
+  fe26_add(nz, zz, zz);    /* 4 */
+  fe26_norm_weak(nz);      /* -> 2 */
   r.x = nx;
   r.y = ny;
   r.z = nz;
@@ -918,61 +1015,8 @@ __device__ KV_DOUBLE_ATTR void gej_double(gej &r, const gej &a) {
   gej_double_impl(r, a);
 }
 
-/* mixed add r = a + B(affine): generic madd formulas computed unconditionally;
- * a==infinity fixed up with selects; the rare equal-x cases (h==0 with a
- * finite) take a divergent slow path only when a lane actually hits them. */
-__device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
-  u64 a_inf = (u64)fe_is_zero(a.z);
-  /* 11 muls as 5 interleaved pairs + 1 */
-  fe z1z1, u2, s2, h, hh, i, j, rr, v, t;
-  fe_sqr_mul(z1z1, a.z, s2, b.y, a.z);
-  fe_mul2(u2, b.x, z1z1, s2, s2, z1z1);
-  fe_sub(h, u2, a.x);
-  fe_sub(rr, s2, a.y);
-  if (!a_inf && fe_is_zero(h)) {
-    /* rare: same x. rr==0 → doubling; else opposite points → infinity */
-    if (fe_is_zero(rr)) {
-      gej_double(r, a);
-    } else {
-      gej_set_infinity(r);
-    }
-    return;
-  }
-  fe zz;
-  fe_add(zz, a.z, h);
-  fe_sqr2(hh, h, zz, zz);
-  fe_add(i, hh, hh);
-  fe_add(i, i, i);
-  fe_mul2(j, h, i, v, a.x, i);
-  fe_add(rr, rr, rr);
-  fe nx, ny, nz;
-  fe y1j;
-  fe_sqr_mul(nx, rr, y1j, a.y, j);
-  fe_sub(nx, nx, j);
-  fe_sub(nx, nx, v);
-  fe_sub(nx, nx, v);
-  fe_sub(t, v, nx);
-  fe_mul(t, rr, t);
-  fe_add(y1j, y1j, y1j);
-  fe_sub(ny, t, y1j);
-  fe_sub(zz, zz, z1z1);
-  fe_sub(zz, zz, hh);
-  nz = zz;
-  /* a was infinity → result is b (z = 1) */
-  static const fe FE_ONE = {{1, 0, 0, 0}};
-  fe_cmov(nx, b.x, a_inf);
-  fe_cmov(ny, b.y, a_inf);
-  fe_cmov(nz, FE_ONE, a_inf);
-  r.x = nx;
-  r.y = ny;
-  r.z = nz;
-}
-
-
-/* four doublings in one call — quarters the ABI crossings of the window ladder */
-/* Four doublings in ONE call frame: the accumulator crosses the noinline ABI
- * (scratch round-trip) once per window instead of four times. The impl is
- * straight-line, so SROA keeps the intermediate points in VGPRs. */
+/* four doublings in one call frame (accumulator crosses the noinline ABI once
+ * per window instead of four times) */
 __device__ KV_GROUP_ATTR void gej_double4(gej &r, const gej &a) {
   gej t;
   gej_double_impl(t, a);
@@ -981,71 +1025,163 @@ __device__ KV_GROUP_ATTR void gej_double4(gej &r, const gej &a) {
   gej_double_impl(r, t);
 }
 
-/* full Jacobian + Jacobian add (add-2007-bl shape), same select discipline as
- * gej_add_ge. b with z==0 produces garbage that callers discard via cmov. */
-__device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
-  u64 a_inf = (u64)fe_is_zero(a.z);
-  /* 16 muls as 8 interleaved pairs */
-  fe z1z1, z2z2, u1, u2, s1, s2, h, i, j, rr, v, t;
-  fe_sqr2(z1z1, a.z, z2z2, b.z);
-  fe_mul2(u1, a.x, z2z2, u2, b.x, z1z1);
-  fe_mul2(s1, a.y, b.z, s2, b.y, a.z);
-  fe_mul2(s1, s1, z2z2, s2, s2, z1z1);
-  fe_sub(h, u2, u1);
-  fe_sub(rr, s2, s1);
-  if (!a_inf && !fe_is_zero(b.z) && fe_is_zero(h)) {
-    if (fe_is_zero(rr)) {
-      gej_double(r, a);
+__device__ __forceinline__ void gej_cmov(gej &r, const gej &a, u64 cond) {
+  u32 c = (u32)(cond & 1);
+  fe26_cmov(r.x, a.x, c);
+  fe26_cmov(r.y, a.y, c);
+  fe26_cmov(r.z, a.z, c);
+}
+
+/* mixed add (b affine, magnitude <= 3 after phi/negation); a coords mag <= 2 */
+__device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
+  u64 a_inf = (u64)fe26_is_zero(a.z);
+  fe26 z1z1, u2, s2, h, hh, i, j, rr, v, t;
+  fe26_sqr(z1z1, a.z);       /* 1 */
+  fe26_mul(s2, b.y, a.z);    /* 1 */
+  fe26_mul(u2, b.x, z1z1);   /* 1 */
+  fe26_mul(s2, s2, z1z1);    /* 1 */
+  fe26 nX, nY;
+  fe26_neg(nX, a.x, 2);      /* 3 */
+  fe26_neg(nY, a.y, 2);      /* 3 */
+  fe26_add(h, u2, nX);       /* 5 */
+  fe26_add(rr, s2, nY);      /* 5 */
+  if (!a_inf && fe26_is_zero(h)) {
+    /* rare: same x. rr==0 -> doubling; else opposite points -> infinity */
+    if (fe26_is_zero(rr)) {
+      gej t2;
+      gej ain = a;
+      fe26_norm_weak(ain.x);
+      fe26_norm_weak(ain.y);
+      fe26_norm_weak(ain.z);
+      gej_double_impl(t2, ain);
+      r = t2;
     } else {
       gej_set_infinity(r);
     }
     return;
   }
-  fe_add(rr, rr, rr); /* r = 2(S2-S1) */
-  fe zz;
-  fe_add(zz, a.z, b.z);
-  fe_sqr2(i, h, zz, zz);
-  fe_add(i, i, i);
-  fe_add(i, i, i); /* I = 4H^2 */
-  fe_mul2(j, h, i, v, u1, i);
-  fe nx, ny, nz;
-  fe s1j;
-  fe_sqr_mul(nx, rr, s1j, s1, j);
-  fe_sub(nx, nx, j);
-  fe_sub(nx, nx, v);
-  fe_sub(nx, nx, v);
-  fe_sub(t, v, nx);
-  fe_sub(zz, zz, z1z1);
-  fe_sub(zz, zz, z2z2);
-  fe_mul2(t, rr, t, nz, zz, h);
-  fe_add(s1j, s1j, s1j);
-  fe_sub(ny, t, s1j);
-  /* a infinity → result = b */
-  fe_cmov(nx, b.x, a_inf);
-  fe_cmov(ny, b.y, a_inf);
-  fe_cmov(nz, b.z, a_inf);
+  fe26 zz;
+  fe26_add(zz, a.z, h);      /* 7 */
+  fe26_sqr(zz, zz);          /* 1 */
+  fe26_sqr(hh, h);           /* 1 */
+  i = hh;
+  fe26_mul_int(i, 4);        /* 8 */
+  fe26_mul(j, h, i);         /* h 5, i 8 -> 1 */
+  fe26_mul(v, a.x, i);       /* 1 */
+  fe26 nx, ny, nz;
+  fe26_sqr(nx, rr);          /* rr 5 -> 1 */
+  fe26_mul_int(nx, 4);       /* 8: (2r)^2 */
+  fe26 nj, nv;
+  fe26_neg(nj, j, 2);        /* 3 */
+  fe26_neg(nv, v, 2);        /* 3 */
+  fe26_add(nx, nx, nj);
+  fe26_add(nx, nx, nv);
+  fe26_add(nx, nx, nv);      /* 8+3+3+3 = 17 */
+  fe26_norm_weak(nx);        /* -> 2 */
+  fe26 nnx;
+  fe26_neg(nnx, nx, 2);      /* 3 */
+  fe26_add(t, v, nnx);       /* 5 */
+  fe26_mul(t, rr, t);        /* rr 5, t 5 -> 1 */
+  fe26_add(t, t, t);         /* 4: 2*rr*(v-nx) */
+  fe26 y1j;
+  fe26_mul(y1j, a.y, j);     /* 1 */
+  fe26_add(y1j, y1j, y1j);   /* 4 */
+  fe26 ny1j;
+  fe26_neg(ny1j, y1j, 4);    /* 5 */
+  fe26_add(ny, t, ny1j);     /* 9 */
+  fe26_norm_weak(ny);        /* -> 2 */
+  fe26 nzz, nhh;
+  fe26_neg(nzz, z1z1, 2);    /* 3 */
+  fe26_neg(nhh, hh, 2);      /* 3 */
+  fe26_add(nz, zz, nzz);
+  fe26_add(nz, nz, nhh);     /* 2+3+3 = 8 */
+  fe26_norm_weak(nz);        /* -> 2 */
+  /* a was infinity -> result is b (z = 1) */
+  fe26 one;
+  fe26_set_int(one, 1);
+  u32 ci = (u32)(a_inf & 1);
+  fe26_cmov(nx, b.x, ci);
+  fe26_cmov(ny, b.y, ci);
+  fe26_cmov(nz, one, ci);
   r.x = nx;
   r.y = ny;
   r.z = nz;
 }
 
-__device__ __forceinline__ void gej_cmov(gej &r, const gej &a, u64 cond) {
-  u64 mask = 0 - cond;
-#pragma unroll
-  for (int i = 0; i < 4; i++) {
-    r.x.n[i] = (r.x.n[i] & ~mask) | (a.x.n[i] & mask);
-    r.y.n[i] = (r.y.n[i] & ~mask) | (a.y.n[i] & mask);
-    r.z.n[i] = (r.z.n[i] & ~mask) | (a.z.n[i] & mask);
+/* full add; a and b coords magnitude <= 3 */
+__device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
+  u64 a_inf = (u64)fe26_is_zero(a.z);
+  fe26 z1z1, z2z2, u1, u2, s1, s2, h, i, j, rr, v, t;
+  fe26_sqr(z1z1, a.z);      /* 1 */
+  fe26_sqr(z2z2, b.z);      /* 1 */
+  fe26_mul(u1, a.x, z2z2);  /* 1 */
+  fe26_mul(u2, b.x, z1z1);  /* 1 */
+  fe26_mul(s1, a.y, b.z);   /* 1 */
+  fe26_mul(s1, s1, z2z2);   /* 1 */
+  fe26_mul(s2, b.y, a.z);   /* 1 */
+  fe26_mul(s2, s2, z1z1);   /* 1 */
+  fe26 nu1, ns1;
+  fe26_neg(nu1, u1, 2);     /* 3 */
+  fe26_neg(ns1, s1, 2);     /* 3 */
+  fe26_add(h, u2, nu1);     /* 5 */
+  fe26_add(rr, s2, ns1);    /* 5 */
+  if (!a_inf && !fe26_is_zero(b.z) && fe26_is_zero(h)) {
+    if (fe26_is_zero(rr)) {
+      gej t2;
+      gej ain = a;
+      fe26_norm_weak(ain.x);
+      fe26_norm_weak(ain.y);
+      fe26_norm_weak(ain.z);
+      gej_double_impl(t2, ain);
+      r = t2;
+    } else {
+      gej_set_infinity(r);
+    }
+    return;
   }
-}
-
-/* branch-light mixed add: computes the generic add formulas unconditionally and
- * fixes up the special cases (a=inf, double, opposite) with selects. The rare
- * truly-special cases (h==0) take a divergent slow path only when they occur. */
-__device__ inline void gej_add_ge_sel(gej &r, const gej &a, const ge &b) {
-  gej out;
-  gej_add_ge(out, a, b);
-  r = out;
+  fe26 zz;
+  fe26_add(zz, a.z, b.z);   /* 2+3 = 5 */
+  fe26_sqr(zz, zz);         /* 1 */
+  fe26_sqr(i, h);           /* 1 */
+  fe26_mul_int(i, 4);       /* 8 */
+  fe26_mul(j, h, i);        /* 1 */
+  fe26_mul(v, u1, i);       /* 1 */
+  fe26 nx, ny, nz;
+  fe26_sqr(nx, rr);         /* 1 */
+  fe26_mul_int(nx, 4);      /* 8 */
+  fe26 nj, nv;
+  fe26_neg(nj, j, 2);
+  fe26_neg(nv, v, 2);
+  fe26_add(nx, nx, nj);
+  fe26_add(nx, nx, nv);
+  fe26_add(nx, nx, nv);     /* 17 */
+  fe26_norm_weak(nx);       /* -> 2 */
+  fe26 nnx;
+  fe26_neg(nnx, nx, 2);
+  fe26_add(t, v, nnx);      /* 5 */
+  fe26_mul(t, rr, t);       /* 1 */
+  fe26_add(t, t, t);        /* 4 */
+  fe26 s1j;
+  fe26_mul(s1j, s1, j);     /* 1 */
+  fe26_add(s1j, s1j, s1j);  /* 4 */
+  fe26 ns1j;
+  fe26_neg(ns1j, s1j, 4);   /* 5 */
+  fe26_add(ny, t, ns1j);    /* 9 */
+  fe26_norm_weak(ny);       /* -> 2 */
+  fe26 nzz1, nzz2;
+  fe26_neg(nzz1, z1z1, 2);
+  fe26_neg(nzz2, z2z2, 2);
+  fe26_add(zz, zz, nzz1);
+  fe26_add(zz, zz, nzz2);   /* 1+3+3 = 7 */
+  fe26_mul(nz, zz, h);      /* zz 7, h 5 -> 1 */
+  /* a infinity -> result = b */
+  u32 ci = (u32)(a_inf & 1);
+  fe26_cmov(nx, b.x, ci);
+  fe26_cmov(ny, b.y, ci);
+  fe26_cmov(nz, b.z, ci);
+  r.x = nx;
+  r.y = ny;
+  r.z = nz;
 }
 
 } // namespace kv

@@ -49,53 +49,47 @@ __device__ __forceinline__ int fe_gte_p_full(const fe &a) {
          (a.n[0] >= KV_P0);
 }
 
-/* lift x (BE bytes) to even-y affine point; 0 on failure */
-__device__ inline int lift_x_even(ge &P, const uint8_t xb[32]) {
-  fe x;
-  fe_from_be(x, xb);
-  if (fe_gte_p_full(x)) return 0;
-  fe x3, y2, y;
-  fe_sqr(x3, x);
-  fe_mul(x3, x3, x);
-  fe seven = {{7, 0, 0, 0}};
-  fe_add(y2, x3, seven);
-  if (!fe_sqrt(y, y2)) return 0;
-  /* even y */
-  fe ny;
-  fe_neg(ny, y);
-  fe_cmov(y, ny, y.n[0] & 1);
+/* lift x (BE bytes) to an affine point with chosen y parity; 0 on failure.
+ * The >= p overflow check runs on the exact 4xu64 form; the curve equation
+ * and sqrt run on fe26. */
+__device__ inline int lift_x_parity(ge &P, const uint8_t xb[32], u32 want_odd) {
+  fe xu;
+  fe_from_be(xu, xb);
+  if (fe_gte_p_full(xu)) return 0;
+  fe26 x, x3, y2, y;
+  fe26_from_fe(x, xu);
+  fe26_sqr(x3, x);
+  fe26_mul(x3, x3, x);
+  fe26 seven;
+  fe26_set_int(seven, 7);
+  fe26_add(y2, x3, seven);
+  if (!fe26_sqrt(y, y2)) return 0;
+  fe26_normalize(y);
+  fe26 ny;
+  fe26_neg(ny, y, 1);
+  fe26_cmov(y, ny, (y.l[0] & 1) ^ want_odd);
   P.x = x;
-  P.y = y;
+  P.y = y; /* magnitude <= 2 */
   return 1;
+}
+
+__device__ inline int lift_x_even(ge &P, const uint8_t xb[32]) {
+  return lift_x_parity(P, xb, 0);
 }
 
 /* parse 33-byte compressed pubkey; 0 on failure */
 __device__ inline int parse_compressed(ge &P, const uint8_t pk[33]) {
   if (pk[0] != 0x02 && pk[0] != 0x03) return 0;
-  fe x;
-  fe_from_be(x, pk + 1);
-  if (fe_gte_p_full(x)) return 0;
-  fe x3, y2, y;
-  fe_sqr(x3, x);
-  fe_mul(x3, x3, x);
-  fe seven = {{7, 0, 0, 0}};
-  fe_add(y2, x3, seven);
-  if (!fe_sqrt(y, y2)) return 0;
-  fe ny;
-  fe_neg(ny, y);
-  u64 want_odd = (pk[0] == 0x03);
-  fe_cmov(y, ny, (y.n[0] & 1) ^ want_odd);
-  P.x = x;
-  P.y = y;
-  return 1;
+  return lift_x_parity(P, pk + 1, pk[0] == 0x03);
 }
 
-/* affine G */
-__device__ __constant__ static const ge GE_G = {
-    {{0x59F2815B16F81798ULL, 0x029BFCDB2DCE28D9ULL, 0x55A06295CE870B07ULL,
-      0x79BE667EF9DCBBACULL}},
-    {{0x9C47D08FFB10D4B8ULL, 0xFD17B448A6855419ULL, 0x5DA4FBFC0E1108A8ULL,
-      0x483ADA7726A3C465ULL}}};
+/* affine G (4xu64 limbs; converted to fe26 once in the table-init kernel) */
+__device__ __constant__ static const u64 GE_G_X[4] = {
+    0x59F2815B16F81798ULL, 0x029BFCDB2DCE28D9ULL, 0x55A06295CE870B07ULL,
+    0x79BE667EF9DCBBACULL};
+__device__ __constant__ static const u64 GE_G_Y[4] = {
+    0x9C47D08FFB10D4B8ULL, 0xFD17B448A6855419ULL, 0x5DA4FBFC0E1108A8ULL,
+    0x483ADA7726A3C465ULL};
 
 /* Precomputed affine multiples 1..15 of G (filled once per device by
  * kv_ec_table_init_kernel; entry 0 unused). Global memory: every lane reads
@@ -104,19 +98,32 @@ __device__ ge KV_G_TABLE[16];
 
 extern "C" __global__ void kv_ec_table_init_kernel() {
   if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  ge G;
+  {
+    fe gx, gy;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      gx.n[i] = GE_G_X[i];
+      gy.n[i] = GE_G_Y[i];
+    }
+    fe26_from_fe(G.x, gx);
+    fe26_from_fe(G.y, gy);
+  }
   gej acc;
-  acc.x = GE_G.x;
-  acc.y = GE_G.y;
-  acc.z = {{1, 0, 0, 0}};
+  acc.x = G.x;
+  acc.y = G.y;
+  fe26_set_int(acc.z, 1);
   for (int k = 1; k <= 15; k++) {
-    fe zi, zi2, zi3;
-    fe_inv(zi, acc.z);
-    fe_sqr(zi2, zi);
-    fe_mul(zi3, zi2, zi);
-    fe_mul(KV_G_TABLE[k].x, acc.x, zi2);
-    fe_mul(KV_G_TABLE[k].y, acc.y, zi3);
+    fe26 zi, zi2, zi3;
+    fe26_inv(zi, acc.z);
+    fe26_sqr(zi2, zi);
+    fe26_mul(zi3, zi2, zi);
+    fe26_mul(KV_G_TABLE[k].x, acc.x, zi2);
+    fe26_mul(KV_G_TABLE[k].y, acc.y, zi3);
+    fe26_normalize(KV_G_TABLE[k].x);
+    fe26_normalize(KV_G_TABLE[k].y);
     gej t;
-    gej_add_ge(t, acc, GE_G);
+    gej_add_ge(t, acc, G);
     acc = t;
   }
 }
@@ -251,13 +258,17 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const g
   gej ptab[16];
   ptab[1].x = P.x;
   ptab[1].y = P.y;
-  ptab[1].z = {{1, 0, 0, 0}};
+  fe26_set_int(ptab[1].z, 1);
 #pragma unroll 1
   for (int k = 2; k <= 15; k++) gej_add_ge(ptab[k], ptab[k - 1], P);
   gej_set_infinity(ptab[0]);
-  fe beta;
+  fe26 beta;
+  {
+    fe bu;
 #pragma unroll
-  for (int i = 0; i < 4; i++) beta.n[i] = GLV_BETA[i];
+    for (int i = 0; i < 4; i++) bu.n[i] = GLV_BETA[i];
+    fe26_from_fe(beta, bu);
+  }
   gej_set_infinity(R);
   if (progress) *progress = 800;
 #pragma unroll 1
@@ -268,9 +279,9 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const g
     {
       u64 d = glv_digit(g1h, w);
       ge e = KV_G_TABLE[d];
-      fe ny;
-      fe_neg(ny, e.y);
-      fe_cmov(e.y, ny, g1h.neg);
+      fe26 ny;
+      fe26_neg(ny, e.y, 2);
+      fe26_cmov(e.y, ny, (u32)g1h.neg);
       gej_add_ge(t, R, e);
       gej_cmov(R, t, (u64)(d != 0));
     }
@@ -278,12 +289,12 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const g
     {
       u64 d = glv_digit(g2h, w);
       ge e = KV_G_TABLE[d];
-      fe bx;
-      fe_mul(bx, e.x, beta);
+      fe26 bx;
+      fe26_mul(bx, e.x, beta);
       e.x = bx;
-      fe ny;
-      fe_neg(ny, e.y);
-      fe_cmov(e.y, ny, g2h.neg);
+      fe26 ny;
+      fe26_neg(ny, e.y, 2);
+      fe26_cmov(e.y, ny, (u32)g2h.neg);
       gej_add_ge(t, R, e);
       gej_cmov(R, t, (u64)(d != 0));
     }
@@ -291,9 +302,9 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const g
     {
       u64 d = glv_digit(p1h, w);
       gej e = ptab[d];
-      fe ny;
-      fe_neg(ny, e.y);
-      fe_cmov(e.y, ny, p1h.neg);
+      fe26 ny;
+      fe26_neg(ny, e.y, 2);
+      fe26_cmov(e.y, ny, (u32)p1h.neg);
       gej_add(t, R, e);
       gej_cmov(R, t, (u64)(d != 0));
     }
@@ -301,12 +312,12 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const g
     {
       u64 d = glv_digit(p2h, w);
       gej e = ptab[d];
-      fe bx;
-      fe_mul(bx, e.x, beta);
+      fe26 bx;
+      fe26_mul(bx, e.x, beta);
       e.x = bx;
-      fe ny;
-      fe_neg(ny, e.y);
-      fe_cmov(e.y, ny, p2h.neg);
+      fe26 ny;
+      fe26_neg(ny, e.y, 2);
+      fe26_cmov(e.y, ny, (u32)p2h.neg);
       gej_add(t, R, e);
       gej_cmov(R, t, (u64)(d != 0));
     }
@@ -318,9 +329,11 @@ __device__ inline uint8_t schnorr_verify_one(const uint8_t *rb, const uint8_t *s
                                              const uint8_t *pkb, const uint8_t *msg) {
   ge P;
   if (!lift_x_even(P, pkb)) return KVS_BAD_PUBKEY;
-  fe rx;
-  fe_from_be(rx, rb);
-  if (fe_gte_p_full(rx)) return KVS_INVALID;
+  fe rxu;
+  fe_from_be(rxu, rb);
+  if (fe_gte_p_full(rxu)) return KVS_INVALID;
+  fe26 rx;
+  fe26_from_fe(rx, rxu);
   sc s;
   if (sc_from_be(s, sb)) return KVS_INVALID;
   uint8_t eh[32];
@@ -332,14 +345,15 @@ __device__ inline uint8_t schnorr_verify_one(const uint8_t *rb, const uint8_t *s
   ecmult_double(R, s, ne, P);
   if (gej_is_infinity(R)) return KVS_INVALID;
   /* affine via one inversion: need x == r and even y */
-  fe zi, zi2, zi3, xa, ya;
-  fe_inv(zi, R.z);
-  fe_sqr(zi2, zi);
-  fe_mul(zi3, zi2, zi);
-  fe_mul(xa, R.x, zi2);
-  fe_mul(ya, R.y, zi3);
-  if (ya.n[0] & 1) return KVS_INVALID;
-  return fe_eq(xa, rx) ? KVS_VALID : KVS_INVALID;
+  fe26 zi, zi2, zi3, xa, ya;
+  fe26_inv(zi, R.z);
+  fe26_sqr(zi2, zi);
+  fe26_mul(zi3, zi2, zi);
+  fe26_mul(xa, R.x, zi2);
+  fe26_mul(ya, R.y, zi3);
+  fe26_normalize(ya);
+  if (ya.l[0] & 1) return KVS_INVALID;
+  return fe26_eq(xa, rx) ? KVS_VALID : KVS_INVALID;
 }
 
 #ifndef KV_LB
@@ -394,18 +408,17 @@ __device__ inline uint8_t ecdsa_verify_one(const uint8_t *rb, const uint8_t *sb,
   ecmult_double(R, u1, u2, P);
   if (gej_is_infinity(R)) return KVS_INVALID;
   /* x(R) ≡ r (mod n): X == (r + k·n)·Z² for k ∈ {0,1} with r+n < p */
-  fe z2;
-  fe_sqr(z2, R.z);
-  uint8_t rbe[32];
-#pragma unroll
-  for (int i = 0; i < 4; i++)
-#pragma unroll
-    for (int j = 0; j < 8; j++) rbe[8 * (3 - i) + j] = (uint8_t)(r.d[i] >> (56 - 8 * j));
+  fe26 z2;
+  fe26_sqr(z2, R.z);
   fe rf;
-  fe_from_be(rf, rbe);
-  fe t;
-  fe_mul(t, rf, z2);
-  if (fe_eq(t, R.x)) return KVS_VALID;
+  rf.n[0] = r.d[0];
+  rf.n[1] = r.d[1];
+  rf.n[2] = r.d[2];
+  rf.n[3] = r.d[3];
+  fe26 rf26, t;
+  fe26_from_fe(rf26, rf);
+  fe26_mul(t, rf26, z2);
+  if (fe26_eq(t, R.x)) return KVS_VALID;
   /* r + n */
   u64 carry = 0;
   fe rn;
@@ -414,8 +427,10 @@ __device__ inline uint8_t ecdsa_verify_one(const uint8_t *rb, const uint8_t *sb,
   rn.n[2] = addc(rf.n[2], KV_N2, carry);
   rn.n[3] = addc(rf.n[3], KV_N3, carry);
   if (!carry && !fe_gte_p_full(rn)) {
-    fe_mul(t, rn, z2);
-    if (fe_eq(t, R.x)) return KVS_VALID;
+    fe26 rn26;
+    fe26_from_fe(rn26, rn);
+    fe26_mul(t, rn26, z2);
+    if (fe26_eq(t, R.x)) return KVS_VALID;
   }
   return KVS_INVALID;
 }
