@@ -68,23 +68,6 @@ __device__ __forceinline__ u64 subb(u64 a, u64 b, u64 &borrow) {
   return (u64)t;
 }
 
-__device__ __forceinline__ int fe_gte_p(const fe &a) {
-  /* a >= p ? branchless */
-  if (a.n[3] != KV_P1 || a.n[2] != KV_P1 || a.n[1] != KV_P1) {
-    return (a.n[3] == KV_P1 && a.n[2] == KV_P1 && a.n[1] == KV_P1) ? (a.n[0] >= KV_P0)
-                                                                   : 0;
-  }
-  return a.n[0] >= KV_P0;
-}
-
-__device__ __forceinline__ void fe_sub_p(fe &a) {
-  u64 borrow = 0;
-  a.n[0] = subb(a.n[0], KV_P0, borrow);
-  a.n[1] = subb(a.n[1], KV_P1, borrow);
-  a.n[2] = subb(a.n[2], KV_P1, borrow);
-  a.n[3] = subb(a.n[3], KV_P1, borrow);
-}
-
 __device__ __forceinline__ void fe_norm_once(fe &a) {
   /* conditional subtract p, branchless */
   u64 ge = (a.n[3] == KV_P1) & (a.n[2] == KV_P1) & (a.n[1] == KV_P1) &
@@ -95,56 +78,6 @@ __device__ __forceinline__ void fe_norm_once(fe &a) {
   a.n[1] = subb(a.n[1], KV_P1 & mask, borrow);
   a.n[2] = subb(a.n[2], KV_P1 & mask, borrow);
   a.n[3] = subb(a.n[3], KV_P1 & mask, borrow);
-}
-
-__device__ __forceinline__ void fe_add(fe &r, const fe &a, const fe &b) {
-  u128 c = (u128)a.n[0] + b.n[0];
-  u64 t0 = (u64)c;
-  c = (c >> 64) + a.n[1] + b.n[1];
-  u64 t1 = (u64)c;
-  c = (c >> 64) + a.n[2] + b.n[2];
-  u64 t2 = (u64)c;
-  c = (c >> 64) + a.n[3] + b.n[3];
-  u64 t3 = (u64)c;
-  /* wrap 2^256 → += PC; a+b < 2^257 so the second wrap adds < 2^34: no ripple
-   * past limb 1 is possible after the first fold unless limbs были max —
-   * handled by a full u128 chain again */
-  u64 w = (u64)(c >> 64) * KV_PC;
-  c = (u128)t0 + w;
-  r.n[0] = (u64)c;
-  c = (c >> 64) + t1;
-  r.n[1] = (u64)c;
-  c = (c >> 64) + t2;
-  r.n[2] = (u64)c;
-  c = (c >> 64) + t3;
-  r.n[3] = (u64)c;
-  if ((u64)(c >> 64)) { /* wrapped again: value is tiny */
-    c = (u128)r.n[0] + KV_PC;
-    r.n[0] = (u64)c;
-    r.n[1] += (u64)(c >> 64);
-  }
-  fe_norm_once(r);
-}
-
-__device__ __forceinline__ void fe_neg(fe &r, const fe &a) {
-  /* p - a for a < p; a==0 → 0 */
-  u64 is_zero = ((a.n[0] | a.n[1] | a.n[2] | a.n[3]) == 0);
-  u64 borrow = 0;
-  r.n[0] = subb(KV_P0, a.n[0], borrow);
-  r.n[1] = subb(KV_P1, a.n[1], borrow);
-  r.n[2] = subb(KV_P1, a.n[2], borrow);
-  r.n[3] = subb(KV_P1, a.n[3], borrow);
-  u64 mask = 0 - is_zero; /* if a==0 result must be 0 not p */
-  r.n[0] &= ~mask;
-  r.n[1] &= ~mask;
-  r.n[2] &= ~mask;
-  r.n[3] &= ~mask;
-}
-
-__device__ __forceinline__ void fe_sub(fe &r, const fe &a, const fe &b) {
-  fe nb;
-  fe_neg(nb, b);
-  fe_add(r, a, nb);
 }
 
 /* full 4x4 multiply (row-wise carry chain) + two-fold reduction via
@@ -201,135 +134,6 @@ __device__ __forceinline__ void fe_mul(fe &r, const fe &a, const fe &b) {
   fe_mul_inner(t, a.n, b.n);
   fe_reduce8(r, t);
 }
-
-/* Two INDEPENDENT multiplies in one straight-line body: the backend scheduler
- * interleaves the two carry chains (separate carry SGPRs), hiding the
- * VALU-carry hazards that serialize a lone fe_mul. Outputs may alias later
- * inputs — both products are fully accumulated before either reduce writes. */
-__device__ __forceinline__ void fe_mul2(fe &r1, const fe &a1, const fe &b1,
-                                        fe &r2, const fe &a2, const fe &b2) {
-  u64 t1[8], t2[8];
-  fe_mul_inner(t1, a1.n, b1.n);
-  fe_mul_inner(t2, a2.n, b2.n);
-  fe_reduce8(r1, t1);
-  fe_reduce8(r2, t2);
-}
-
-/* dedicated squaring: 6 cross + 4 diagonal 64×64 products (vs 16 for the
- * generic multiply) — ~48% of the ladder's field muls are squarings. Cross
- * products accumulate un-doubled (every step adds ≤ product + 2 u64, safely
- * inside u128), then one shift-chain doubles them before the diagonals. */
-__device__ __forceinline__ void fe_sqr_inner(u64 t[8], const u64 *a) {
-  u64 t1, t2, t3, t4, t5, t6, c;
-  u128 m;
-  m = (u128)a[0] * a[1];
-  t1 = (u64)m;
-  c = (u64)(m >> 64);
-  m = (u128)a[0] * a[2] + c;
-  t2 = (u64)m;
-  c = (u64)(m >> 64);
-  m = (u128)a[0] * a[3] + c;
-  t3 = (u64)m;
-  t4 = (u64)(m >> 64);
-  m = (u128)a[1] * a[2] + t3;
-  t3 = (u64)m;
-  c = (u64)(m >> 64);
-  m = (u128)a[1] * a[3] + t4 + c;
-  t4 = (u64)m;
-  t5 = (u64)(m >> 64);
-  m = (u128)a[2] * a[3] + t5;
-  t5 = (u64)m;
-  t6 = (u64)(m >> 64);
-  /* double the cross part */
-  u64 t7 = t6 >> 63;
-  t6 = (t6 << 1) | (t5 >> 63);
-  t5 = (t5 << 1) | (t4 >> 63);
-  t4 = (t4 << 1) | (t3 >> 63);
-  t3 = (t3 << 1) | (t2 >> 63);
-  t2 = (t2 << 1) | (t1 >> 63);
-  t1 <<= 1;
-  /* diagonals into even columns */
-  u64 lo0, hi0, lo1, hi1, lo2, hi2, lo3, hi3;
-  m = (u128)a[0] * a[0];
-  lo0 = (u64)m;
-  hi0 = (u64)(m >> 64);
-  m = (u128)a[1] * a[1];
-  lo1 = (u64)m;
-  hi1 = (u64)(m >> 64);
-  m = (u128)a[2] * a[2];
-  lo2 = (u64)m;
-  hi2 = (u64)(m >> 64);
-  m = (u128)a[3] * a[3];
-  lo3 = (u64)m;
-  hi3 = (u64)(m >> 64);
-  t[0] = lo0;
-  m = (u128)t1 + hi0;
-  t[1] = (u64)m;
-  m = (m >> 64) + t2 + lo1;
-  t[2] = (u64)m;
-  m = (m >> 64) + t3 + hi1;
-  t[3] = (u64)m;
-  m = (m >> 64) + t4 + lo2;
-  t[4] = (u64)m;
-  m = (m >> 64) + t5 + hi2;
-  t[5] = (u64)m;
-  m = (m >> 64) + t6 + lo3;
-  t[6] = (u64)m;
-  m = (m >> 64) + t7 + hi3;
-  t[7] = (u64)m;
-}
-
-__device__ __forceinline__ void fe_sqr(fe &r, const fe &a) {
-  u64 t[8];
-  fe_sqr_inner(t, a.n);
-  fe_reduce8(r, t);
-}
-
-/* paired variants (same interleaving rationale as fe_mul2) */
-__device__ __forceinline__ void fe_sqr2(fe &r1, const fe &a1, fe &r2,
-                                        const fe &a2) {
-  u64 t1[8], t2[8];
-  fe_sqr_inner(t1, a1.n);
-  fe_sqr_inner(t2, a2.n);
-  fe_reduce8(r1, t1);
-  fe_reduce8(r2, t2);
-}
-
-__device__ __forceinline__ void fe_sqr_mul(fe &rs, const fe &as, fe &rm,
-                                           const fe &am, const fe &bm) {
-  u64 t1[8], t2[8];
-  fe_sqr_inner(t1, as.n);
-  fe_mul_inner(t2, am.n, bm.n);
-  fe_reduce8(rs, t1);
-  fe_reduce8(rm, t2);
-}
-
-__device__ __forceinline__ void fe_mul_small(fe &r, const fe &a, u64 k) {
-  u128 c = (u128)a.n[0] * k;
-  u64 t0 = (u64)c;
-  c = (c >> 64) + (u128)a.n[1] * k;
-  u64 t1 = (u64)c;
-  c = (c >> 64) + (u128)a.n[2] * k;
-  u64 t2 = (u64)c;
-  c = (c >> 64) + (u128)a.n[3] * k;
-  u64 t3 = (u64)c;
-  u64 t4 = (u64)(c >> 64);
-  c = (u128)t0 + (u128)t4 * KV_PC;
-  r.n[0] = (u64)c;
-  c = (c >> 64) + t1;
-  r.n[1] = (u64)c;
-  c = (c >> 64) + t2;
-  r.n[2] = (u64)c;
-  c = (c >> 64) + t3;
-  r.n[3] = (u64)c;
-  if ((u64)(c >> 64)) {
-    c = (u128)r.n[0] + KV_PC;
-    r.n[0] = (u64)c;
-    r.n[1] += (u64)(c >> 64);
-  }
-  fe_norm_once(r);
-}
-
 
 /* ================= 10x26 field representation (fe26) =================
  * (limb convention: l0..l8 < m*2^26, l9 < m*2^22 for magnitude m)
@@ -581,94 +385,6 @@ __device__ __forceinline__ void fe26_to_fe(fe &r, const fe26 &a) {
   r.n[2] = ((u64)a.l[4] >> 24) | ((u64)a.l[5] << 2) | ((u64)a.l[6] << 28) |
            ((u64)a.l[7] << 54);
   r.n[3] = ((u64)a.l[7] >> 10) | ((u64)a.l[8] << 16) | ((u64)a.l[9] << 42);
-}
-
-__device__ __forceinline__ int fe_is_zero(const fe &a) {
-  return (a.n[0] | a.n[1] | a.n[2] | a.n[3]) == 0;
-}
-
-__device__ __forceinline__ int fe_eq(const fe &a, const fe &b) {
-  return ((a.n[0] ^ b.n[0]) | (a.n[1] ^ b.n[1]) | (a.n[2] ^ b.n[2]) |
-          (a.n[3] ^ b.n[3])) == 0;
-}
-
-__device__ __forceinline__ void fe_cmov(fe &r, const fe &a, u64 cond) {
-  u64 mask = 0 - cond;
-#pragma unroll
-  for (int i = 0; i < 4; i++) r.n[i] = (r.n[i] & ~mask) | (a.n[i] & mask);
-}
-
-__device__ KV_GROUP_ATTR void fe_sqrn(fe &r, int n) {
-#pragma unroll 1
-  for (int i = 0; i < n; i++) fe_sqr(r, r);
-}
-
-/* Addition-chain blocks a^(2^k-1) shared by the p-2 (inverse) and (p+1)/4
- * (sqrt) exponents — both exponents have 2^223-1 as their high part because
- * p = 2^256 - 2^32 - 977. The chain is derived and verified symbolically (the
- * exponent algebra is checked in python against p-2 and (p+1)/4 before
- * transcription); it replaces a naive 256-step square-and-multiply (≈512
- * fe_mul) with ≈269. noinline: part of the gfx950 long-branch discipline. */
-__device__ KV_GROUP_ATTR void fe_chain223(fe &x223, fe &x22, fe &x2, fe &x3,
-                                          const fe &a) {
-  fe t, x11, x44, x88;
-  fe_sqr(t, a);
-  fe_mul(x2, t, a); /* 2^2-1 */
-  fe_sqr(t, x2);
-  fe_mul(x3, t, a); /* 2^3-1 */
-  t = x3;
-  fe_sqrn(t, 3);
-  fe_mul(t, t, x3); /* x6 */
-  fe_sqrn(t, 3);
-  fe_mul(t, t, x3); /* x9 */
-  fe_sqrn(t, 2);
-  fe_mul(x11, t, x2);
-  t = x11;
-  fe_sqrn(t, 11);
-  fe_mul(x22, t, x11);
-  t = x22;
-  fe_sqrn(t, 22);
-  fe_mul(x44, t, x22);
-  t = x44;
-  fe_sqrn(t, 44);
-  fe_mul(x88, t, x44);
-  t = x88;
-  fe_sqrn(t, 88);
-  fe_mul(t, t, x88); /* x176 */
-  fe_sqrn(t, 44);
-  fe_mul(t, t, x44); /* x220 */
-  fe_sqrn(t, 3);
-  fe_mul(x223, t, x3);
-}
-
-/* a^(p-2): 255 sqr + 14 mul */
-__device__ inline void fe_inv(fe &r, const fe &a) {
-  fe x223, x22, x2, x3, t;
-  fe_chain223(x223, x22, x2, x3, a);
-  t = x223;
-  fe_sqrn(t, 23);
-  fe_mul(t, t, x22);
-  fe_sqrn(t, 5);
-  fe_mul(t, t, a);
-  fe_sqrn(t, 3);
-  fe_mul(t, t, x2);
-  fe_sqrn(t, 2);
-  fe_mul(r, t, a);
-}
-
-/* sqrt via a^((p+1)/4), 253 sqr + 13 mul; returns 1 if r*r == a */
-__device__ inline int fe_sqrt(fe &r, const fe &a) {
-  fe x223, x22, x2, x3, t, chk;
-  fe_chain223(x223, x22, x2, x3, a);
-  t = x223;
-  fe_sqrn(t, 23);
-  fe_mul(t, t, x22);
-  fe_sqrn(t, 6);
-  fe_mul(t, t, x2);
-  fe_sqrn(t, 2);
-  fe_sqr(chk, t);
-  r = t;
-  return fe_eq(chk, a);
 }
 
 /* ---------- scalar mod n ---------- */
