@@ -104,7 +104,9 @@ def block_mode(args):
     log(f"generated {n_txs} mixed txs ({len(blob)/1e6:.1f} MB blob) "
         f"in {time.time()-t0:.1f}s")
     from rusty_kaspa_amd.engine import Engine
-    eng = Engine()
+    # sig cache OFF for the bench: the batch is reused across timed steps and
+    # a cache hit would skip the verify work being measured
+    eng = Engine(sig_cache_size=0)
     n_inputs = sum(1 for _ in range(0))  # informational only
 
     def one_step():
@@ -150,7 +152,7 @@ def block_mode(args):
         "data": "synthetic (seeded oracle-signed config-3 mix; batch reused across steps)",
         "config": {"workload": "block-validate-config3",
                    "blocks_per_step": args.block_batch, "txs_per_block": 300,
-                   "mix": "70p2pk/20multi-in/10ecdsa", "flags": "FULL",
+                   "mix": "70p2pk/20multi-in/10ecdsa", "flags": "FULL", "sig_cache": "off",
                    "parallelism": "single"},
         "roofline": None,  # per-kernel rooflines live in the default mode + profiles/
         "cpu_baseline": cpu_baseline,
@@ -192,7 +194,7 @@ def main():
     tuples = gen_tuples(oracle, n, SEED + rank, args.invalid_permille)
 
     from rusty_kaspa_amd.engine import Engine
-    eng = Engine(device=local_rank)
+    eng = Engine(device=local_rank, sig_cache_size=0)
     lib = eng.lib
     ctx = ctypes.c_void_p(eng.ctx)
 

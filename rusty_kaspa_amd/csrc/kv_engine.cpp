@@ -17,6 +17,8 @@
 #include <stdlib.h>
 #include <string.h>
 #include <string>
+#include <array>
+#include <unordered_map>
 
 /* kernels compiled into this TU (single translation unit keeps the build to one
  * hipcc invocation, no -fgpu-rdc) */
@@ -40,6 +42,17 @@ extern "C" const char *kv_last_error(void) { return g_last_error.c_str(); }
     }                                                                          \
   } while (0)
 
+struct kv_sig_key {
+  uint64_t w[4];
+  bool operator==(const kv_sig_key &o) const {
+    return ((w[0] ^ o.w[0]) | (w[1] ^ o.w[1]) | (w[2] ^ o.w[2]) |
+            (w[3] ^ o.w[3])) == 0;
+  }
+};
+struct kv_sig_key_hash {
+  size_t operator()(const kv_sig_key &k) const { return (size_t)k.w[0]; }
+};
+
 struct kv_ctx {
   kv_params params;
   hipStream_t stream;
@@ -52,6 +65,11 @@ struct kv_ctx {
   uint8_t *d_status = nullptr;
   size_t d_status_cap = 0;
   uint64_t cache_hits = 0, cache_misses = 0, cache_insertions = 0;
+  /* sig cache ⇔ TransactionValidator sig_cache (crypto/txscript/src/caches.rs:
+   * 57-82): verdicts of (txid, entries-digest, input, hash-type, sig, pk)
+   * keyed checks survive across calls — the mempool→block revalidation dedup.
+   * Key is a blake2b-256 so a collision is cryptographically excluded. */
+  std::unordered_map<kv_sig_key, uint8_t, kv_sig_key_hash> sig_cache;
   /* GPU-resident UTXO set */
   kv::utxo_slot *d_utxo = nullptr;
   uint64_t utxo_cap = 0; /* power of two */
@@ -932,9 +950,93 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
       }
   }
 
-  /* phase 2: GPU — subhashes, sighash+tuple assembly, EC verify */
+  /* phase 1.5: sig cache ⇔ TransactionValidator sig_cache (caches.rs:57-82).
+   * Key = blake2b-256(tx_id ‖ digest(all input entries) ‖ input_index ‖
+   * hash_type ‖ ecdsa ‖ sig ‖ pk): tx_id pins everything else the sighash
+   * commits to (it excludes sig scripts and utxo entries — those are keyed
+   * explicitly). Hits skip the GPU entirely; the job lists are compacted to
+   * the misses and statuses scattered back. */
+  size_t ns_all = sjobs.size(), ne_all = ejobs.size();
+  std::vector<uint8_t> s_status(ns_all), e_status(ne_all);
+  std::vector<uint32_t> s_map, e_map;
+  std::vector<kv_sig_key> s_keys, e_keys;
+  bool use_cache = ctx->params.sig_cache_size > 0 && ns_all + ne_all > 0;
+  if (use_cache) {
+    std::vector<uint8_t> need(n_txs, 0);
+    for (auto &j : sjobs) need[j.tx_index] = 1;
+    for (auto &j : ejobs) need[j.tx_index] = 1;
+    std::vector<std::array<uint8_t, 32>> ed(n_txs);
+    kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
+      if (!need[t]) return;
+      std::vector<uint8_t> buf;
+      buf.reserve(txs[t].inputs.size() * 60);
+      for (auto &in : txs[t].inputs) {
+        uint8_t tmp[22];
+        memcpy(tmp, &in.utxo_amount, 8);
+        memcpy(tmp + 8, &in.utxo_daa_score, 8);
+        tmp[16] = in.utxo_is_coinbase;
+        tmp[17] = in.utxo_has_cov;
+        memcpy(tmp + 18, &in.utxo_spk_version, 2);
+        uint16_t sl = (uint16_t)in.utxo_spk_len;
+        memcpy(tmp + 20, &sl, 2);
+        buf.insert(buf.end(), tmp, tmp + 22);
+        buf.insert(buf.end(), in.utxo_spk, in.utxo_spk + in.utxo_spk_len);
+      }
+      h_blake2b_keyed(nullptr, 0, buf.data(), buf.size(), ed[t].data());
+    });
+    auto keygen = [&](const std::vector<kv::kv_job> &jobs,
+                      std::vector<kv_sig_key> &keys) {
+      keys.resize(jobs.size());
+      kvh_parallel_for((uint32_t)jobs.size(), [&](uint32_t i) {
+        const kv::kv_job &j = jobs[i];
+        uint8_t buf[32 + 32 + 4 + 2 + 64 + 33];
+        size_t o = 0;
+        memcpy(buf + o, txs[j.tx_index].tx_id, 32);
+        o += 32;
+        memcpy(buf + o, ed[j.tx_index].data(), 32);
+        o += 32;
+        memcpy(buf + o, &j.input_index, 4);
+        o += 4;
+        buf[o++] = j.hash_type;
+        buf[o++] = j.ecdsa;
+        memcpy(buf + o, blob + j.sig_off, 64);
+        o += 64;
+        size_t pklen = j.ecdsa ? 33 : 32;
+        memcpy(buf + o, blob + j.pk_off, pklen);
+        o += pklen;
+        uint8_t h[32];
+        h_blake2b_keyed(nullptr, 0, buf, o, h);
+        memcpy(keys[i].w, h, 32);
+      });
+    };
+    keygen(sjobs, s_keys);
+    keygen(ejobs, e_keys);
+    auto probe = [&](std::vector<kv::kv_job> &jobs,
+                     const std::vector<kv_sig_key> &keys,
+                     std::vector<uint8_t> &status, std::vector<uint32_t> &map) {
+      std::vector<kv::kv_job> miss;
+      miss.reserve(jobs.size());
+      for (size_t i = 0; i < jobs.size(); i++) {
+        auto it = ctx->sig_cache.find(keys[i]);
+        if (it != ctx->sig_cache.end()) {
+          status[i] = it->second;
+          ctx->cache_hits++;
+        } else {
+          ctx->cache_misses++;
+          map.push_back((uint32_t)i);
+          miss.push_back(jobs[i]);
+        }
+      }
+      jobs.swap(miss);
+    };
+    probe(sjobs, s_keys, s_status, s_map);
+    probe(ejobs, e_keys, e_status, e_map);
+  }
+
+  /* phase 2: GPU — subhashes, sighash+tuple assembly, EC verify (cache
+   * misses only when the cache is on) */
   size_t ns = sjobs.size(), ne = ejobs.size();
-  std::vector<uint8_t> s_status(ns), e_status(ne);
+  std::vector<uint8_t> s_gpu(ns), e_gpu(ne);
   bool blob_uploaded = false;
   if (ns + ne > 0) {
     if (g_vb.blob.ensure(blob_len) || g_vb.subhashes.ensure((size_t)n_txs * 160))
@@ -961,7 +1063,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
                          dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.s_tuples.p,
                          (unsigned long long)ns, (unsigned long long *)g_vb.s_bitmap.p,
                          (uint8_t *)g_vb.s_status.p);
-      HIP_CHECK(hipMemcpyAsync(s_status.data(), g_vb.s_status.p, ns,
+      HIP_CHECK(hipMemcpyAsync(s_gpu.data(), g_vb.s_status.p, ns,
                                hipMemcpyDeviceToHost, ctx->stream));
     }
     if (ne) {
@@ -980,11 +1082,33 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
                          dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.e_tuples.p,
                          (unsigned long long)ne, (unsigned long long *)g_vb.e_bitmap.p,
                          (uint8_t *)g_vb.e_status.p);
-      HIP_CHECK(hipMemcpyAsync(e_status.data(), g_vb.e_status.p, ne,
+      HIP_CHECK(hipMemcpyAsync(e_gpu.data(), g_vb.e_status.p, ne,
                                hipMemcpyDeviceToHost, ctx->stream));
     }
     HIP_CHECK(hipGetLastError());
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  }
+
+  /* scatter GPU statuses back and remember fresh verdicts */
+  if (use_cache) {
+    size_t cap = (size_t)ctx->params.sig_cache_size;
+    for (size_t k = 0; k < ns; k++) {
+      s_status[s_map[k]] = s_gpu[k];
+      if (ctx->sig_cache.size() >= cap)
+        ctx->sig_cache.erase(ctx->sig_cache.begin());
+      ctx->sig_cache.emplace(s_keys[s_map[k]], s_gpu[k]);
+      ctx->cache_insertions++;
+    }
+    for (size_t k = 0; k < ne; k++) {
+      e_status[e_map[k]] = e_gpu[k];
+      if (ctx->sig_cache.size() >= cap)
+        ctx->sig_cache.erase(ctx->sig_cache.begin());
+      ctx->sig_cache.emplace(e_keys[e_map[k]], e_gpu[k]);
+      ctx->cache_insertions++;
+    }
+  } else {
+    s_status.swap(s_gpu);
+    e_status.swap(e_gpu);
   }
 
   /* phase 3: resolution (first failing input wins, sequential semantics);

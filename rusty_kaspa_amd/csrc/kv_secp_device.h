@@ -230,6 +230,7 @@ __device__ __forceinline__ void fe26_mul_inner(u64 t[19], const fe26 &a,
   for (int i = 0; i < 10; i++)
 #pragma unroll
     for (int j = 0; j < 10; j++) t[i + j] += (u64)a.l[i] * b.l[j];
+
 }
 
 __device__ __forceinline__ void fe26_sqr_inner(u64 t[19], const fe26 &a) {

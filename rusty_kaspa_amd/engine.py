@@ -42,9 +42,10 @@ class Engine:
     consensus/src/processes/transaction_validator/mod.rs:15)."""
 
     def __init__(self, device: int = -1, coinbase_maturity: int = 1000,
-                 mass_per_sig_op: int = 1000):
+                 mass_per_sig_op: int = 1000, sig_cache_size: int = 10_000):
         self.lib = load_library()
-        params = KvParams(coinbase_maturity, mass_per_sig_op, 10_000, device)
+        params = KvParams(coinbase_maturity, mass_per_sig_op, sig_cache_size,
+                          device)
         self.ctx = self.lib.kv_create(ctypes.byref(params))
         if not self.ctx:
             err = self.lib.kv_last_error().decode()
@@ -129,3 +130,12 @@ class Engine:
             codes, fees, partial)
         self._check(rc)
         return list(codes), list(fees), bytes(partial) if want_muhash else None
+
+    def sig_cache_stats(self):
+        class _CS(ctypes.Structure):
+            _fields_ = [("insertions", ctypes.c_uint64), ("hits", ctypes.c_uint64),
+                        ("misses", ctypes.c_uint64)]
+        out = _CS()
+        self._check(self.lib.kv_sig_cache_stats(ctypes.c_void_p(self.ctx),
+                                                ctypes.byref(out)))
+        return out.insertions, out.hits, out.misses
