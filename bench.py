@@ -109,8 +109,29 @@ def block_mode(args):
     eng = Engine(sig_cache_size=0)
     n_inputs = sum(1 for _ in range(0))  # informational only
 
+    utxo = args.mode == "block-utxo"
+    if utxo:
+        # populate path: entries resolve from the GPU-resident UTXO table
+        # (kv_validate_block_utxo). apply_diff stays off so the step is
+        # steady-state repeatable; the diff-apply cost is covered by
+        # tests/test_gpu_validate_utxo.py.
+        from rusty_kaspa_amd.blob import strip_utxo_entries
+        stripped, seeds = strip_utxo_entries(blob)
+        lib = eng.lib
+        ctx = ctypes.c_void_p(eng.ctx)
+        assert lib.kv_utxo_reset(ctx, ctypes.c_uint64(2 * len(seeds))) == 0
+        ops = b"".join(op for op, _ in seeds)
+        ents = b"".join(e for _, e in seeds)
+        assert lib.kv_utxo_upsert(ctx, ops, ents, ctypes.c_size_t(len(seeds))) == 0
+        log(f"UTXO table seeded: {len(seeds)} entries")
+        blob = stripped
+
     def one_step():
-        codes, fees, partial = eng.validate_block(blob, n_txs, 10**9, 10**9, 0)
+        if utxo:
+            codes, fees, partial = eng.validate_block_utxo(
+                blob, n_txs, 10**9, 10**9, 0, apply_diff=False)
+        else:
+            codes, fees, partial = eng.validate_block(blob, n_txs, 10**9, 10**9, 0)
         mh = eng.muhash_finalize(partial)
         return codes, mh
 
@@ -150,7 +171,8 @@ def block_mode(args):
         "vs_baseline": None,
         "dtype": "u256",
         "data": "synthetic (seeded oracle-signed config-3 mix; batch reused across steps)",
-        "config": {"workload": "block-validate-config3",
+        "config": {"workload": ("block-validate-utxo-config3" if args.mode == "block-utxo"
+                                else "block-validate-config3"),
                    "blocks_per_step": args.block_batch, "txs_per_block": 300,
                    "mix": "70p2pk/20multi-in/10ecdsa", "flags": "FULL", "sig_cache": "off",
                    "parallelism": "single"},
@@ -169,12 +191,13 @@ def main():
     ap.add_argument("--tuples", type=int, default=DEFAULT_TUPLES)
     ap.add_argument("--invalid-permille", type=int, default=INVALID_PERMILLE)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--mode", choices=["verify", "block"], default="verify")
+    ap.add_argument("--mode", choices=["verify", "block", "block-utxo"],
+                    default="verify")
     ap.add_argument("--block-batch", type=int, default=16,
                     help="blocks (of 300 txs) per step in --mode block")
     args = ap.parse_args()
 
-    if args.mode == "block":
+    if args.mode in ("block", "block-utxo"):
         return block_mode(args)
 
     rank = int(os.environ.get("RANK", "0"))
