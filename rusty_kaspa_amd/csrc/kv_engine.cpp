@@ -1391,17 +1391,21 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
     return -1;
   }
 
-  /* populate: one GPU lookup over every input's outpoint */
-  std::vector<uint8_t> ops;
-  size_t n_in_total = 0;
-  for (auto &tx : txs) n_in_total += tx.inputs.size();
-  ops.reserve(n_in_total * 36);
-  for (auto &tx : txs)
-    for (auto &in : tx.inputs) {
-      ops.insert(ops.end(), in.prev_tx_id, in.prev_tx_id + 32);
-      const uint8_t *ix = (const uint8_t *)&in.prev_index;
-      ops.insert(ops.end(), ix, ix + 4);
+  /* populate: one GPU lookup over every input's outpoint. The gather and the
+   * blob rebuild fan over the host pool (per-tx prefix-summed offsets). */
+  std::vector<size_t> in_base(n_txs + 1, 0);
+  for (int t = 0; t < n_txs; t++)
+    in_base[t + 1] = in_base[t] + txs[t].inputs.size();
+  size_t n_in_total = in_base[n_txs];
+  std::vector<uint8_t> ops(n_in_total * 36);
+  kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
+    uint8_t *dst = ops.data() + in_base[t] * 36;
+    for (auto &in : txs[t].inputs) {
+      memcpy(dst, in.prev_tx_id, 32);
+      memcpy(dst + 32, &in.prev_index, 4);
+      dst += 36;
     }
+  });
   std::vector<uint8_t> entries(n_in_total * 64);
   std::vector<uint64_t> found((n_in_total + 63) / 64, 0);
   if (n_in_total) {
@@ -1412,31 +1416,54 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
 
   /* rebuild the blob with populated entries; pre-fail txs with missing inputs */
   std::vector<int32_t> pre_codes(n_txs, 0);
-  std::vector<uint8_t> pop;
-  pop.reserve(blob_len + n_in_total * 40);
-  pop.resize(4 + 4ull * n_txs);
-  memcpy(pop.data(), blob, 4);
-  size_t in_idx = 0;
-  for (int t = 0; t < n_txs; t++) {
+  std::vector<size_t> new_off(n_txs + 1, 0);
+  kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
     const HTx &tx = txs[t];
-    uint32_t new_off = (uint32_t)pop.size();
-    memcpy(pop.data() + 4 + 4ull * t, &new_off, 4);
-    uint32_t tx_end = (t + 1 < n_txs) ? txs[t + 1].off : (uint32_t)blob_len;
-    if (tx.inputs.empty()) { /* coinbase or inputless: copy verbatim */
-      pop.insert(pop.end(), blob + tx.off, blob + tx_end);
-      continue;
+    uint32_t tx_end = ((int)t + 1 < n_txs) ? txs[t + 1].off : (uint32_t)blob_len;
+    size_t sz;
+    if (tx.inputs.empty()) {
+      sz = tx_end - tx.off;
+    } else {
+      sz = tx.inputs[0].rec_off - tx.off; /* header + payload */
+      size_t ii = in_base[t];
+      for (auto &in : tx.inputs) {
+        int hit = (found[ii / 64] >> (ii % 64)) & 1;
+        uint32_t spk_len = 0;
+        if (hit) memcpy(&spk_len, entries.data() + ii * 64 + 20, 4);
+        else if (!pre_codes[t]) pre_codes[t] = KV_ERR_MISSING_OUTPOINT;
+        sz += 52 + in.sig_script_len + 24 + spk_len;
+        ii++;
+      }
+      uint32_t outs_start = tx.outputs.empty() ? tx_end : tx.output_offs[0];
+      sz += tx_end - outs_start;
     }
-    /* header + payload */
-    pop.insert(pop.end(), blob + tx.off, blob + tx.inputs[0].rec_off);
-    for (size_t i = 0; i < tx.inputs.size(); i++, in_idx++) {
-      const HInput &in = tx.inputs[i];
-      /* outpoint/sequence/commit/sig_script prefix, verbatim */
-      pop.insert(pop.end(), blob + in.rec_off,
-                 blob + in.rec_off + 52 + in.sig_script_len);
-      int hit = (found[in_idx / 64] >> (in_idx % 64)) & 1;
-      if (!hit && !pre_codes[t]) pre_codes[t] = KV_ERR_MISSING_OUTPOINT;
-      /* entry record (64B packed, kv_utxo_kernels.hip layout) → blob fields */
-      const uint8_t *e = entries.data() + in_idx * 64;
+    new_off[t + 1] = sz; /* size for now; prefixed below */
+  });
+  size_t hdr_len = 4 + 4ull * n_txs;
+  new_off[0] = hdr_len;
+  for (int t = 0; t < n_txs; t++) new_off[t + 1] += new_off[t];
+  std::vector<uint8_t> pop(new_off[n_txs]);
+  memcpy(pop.data(), blob, 4);
+  kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
+    const HTx &tx = txs[t];
+    uint32_t off32 = (uint32_t)new_off[t];
+    memcpy(pop.data() + 4 + 4ull * t, &off32, 4);
+    uint8_t *dst = pop.data() + new_off[t];
+    uint32_t tx_end = ((int)t + 1 < n_txs) ? txs[t + 1].off : (uint32_t)blob_len;
+    if (tx.inputs.empty()) {
+      memcpy(dst, blob + tx.off, tx_end - tx.off);
+      return;
+    }
+    size_t hp = tx.inputs[0].rec_off - tx.off;
+    memcpy(dst, blob + tx.off, hp);
+    dst += hp;
+    size_t ii = in_base[t];
+    for (auto &in : tx.inputs) {
+      size_t pre = 52 + in.sig_script_len;
+      memcpy(dst, blob + in.rec_off, pre);
+      dst += pre;
+      int hit = (found[ii / 64] >> (ii % 64)) & 1;
+      const uint8_t *e = entries.data() + ii * 64;
       uint64_t amount = 0, daa = 0;
       uint16_t eflags = 0, spkv = 0;
       uint32_t spk_len = 0;
@@ -1447,22 +1474,22 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
         memcpy(&spkv, e + 18, 2);
         memcpy(&spk_len, e + 20, 4);
       }
-      uint8_t hdr[24];
-      memcpy(hdr, &amount, 8);
-      memcpy(hdr + 8, &daa, 8);
-      hdr[16] = (uint8_t)(eflags & 1);
-      hdr[17] = 0; /* covenants out of round-1 scope */
-      memcpy(hdr + 18, &spkv, 2);
-      memcpy(hdr + 20, &spk_len, 4);
-      pop.insert(pop.end(), hdr, hdr + 24);
-      if (spk_len) pop.insert(pop.end(), e + 24, e + 24 + spk_len);
+      memcpy(dst, &amount, 8);
+      memcpy(dst + 8, &daa, 8);
+      dst[16] = (uint8_t)(eflags & 1);
+      dst[17] = 0; /* covenants out of round-1 scope */
+      memcpy(dst + 18, &spkv, 2);
+      memcpy(dst + 20, &spk_len, 4);
+      dst += 24;
+      if (spk_len) {
+        memcpy(dst, e + 24, spk_len);
+        dst += spk_len;
+      }
+      ii++;
     }
-    /* outputs region runs to the end of the tx */
-    uint32_t outs_start = tx.outputs.empty()
-                              ? tx_end
-                              : tx.output_offs[0];
-    pop.insert(pop.end(), blob + outs_start, blob + tx_end);
-  }
+    uint32_t outs_start = tx.outputs.empty() ? tx_end : tx.output_offs[0];
+    memcpy(dst, blob + outs_start, tx_end - outs_start);
+  });
 
   int rc = validate_block_impl(ctx, pop.data(), pop.size(), pov_daa_score,
                                block_daa_score, flags, tx_codes_out, fees_out,
@@ -1471,18 +1498,13 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
 
   /* diff apply for accepted txs: remove spent, upsert created */
   std::vector<uint8_t> del_ops, add_ops, add_ents;
-  in_idx = 0;
   for (int t = 0; t < n_txs; t++) {
     const HTx &tx = txs[t];
-    if (tx_codes_out[t] != 0) {
-      in_idx += tx.inputs.size();
-      continue;
-    }
-    for (auto &in : tx.inputs) {
+    if (tx_codes_out[t] != 0) continue;
+    size_t in_idx = in_base[t];
+    for (size_t i = 0; i < tx.inputs.size(); i++, in_idx++)
       del_ops.insert(del_ops.end(), ops.begin() + in_idx * 36,
                      ops.begin() + (in_idx + 1) * 36);
-      in_idx++;
-    }
     for (uint32_t i = 0; i < tx.outputs.size(); i++) {
       const HOutput &o = tx.outputs[i];
       if (o.spk_len > 36) {
