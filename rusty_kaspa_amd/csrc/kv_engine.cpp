@@ -18,6 +18,7 @@
 #include <string.h>
 #include <string>
 #include <array>
+#include <chrono>
 #include <unordered_map>
 
 /* kernels compiled into this TU (single translation unit keeps the build to one
@@ -1319,8 +1320,12 @@ static int utxo_lookup_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
     set_error("kv_utxo_lookup: call kv_utxo_reset first");
     return -1;
   }
+  const bool kvt = getenv("KV_TIMING") != nullptr;
+  auto tn = []() { return std::chrono::steady_clock::now(); };
+  auto t0_ = tn();
   int rc = utxo_stage(ctx, outpoints, nullptr, n);
   if (rc) return rc;
+  auto t1_ = tn();
   size_t words = (n + 63) / 64;
   if (ensure_cap((void **)&ctx->d_ent_out, &ctx->d_ent_cap, n * 64)) return -2;
   if (ensure_cap((void **)&ctx->d_bitmap, &ctx->d_bitmap_cap, words * 8)) return -2;
@@ -1339,7 +1344,13 @@ static int utxo_lookup_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
                              hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipMemcpyAsync(found_bitmap, ctx->d_bitmap, words * 8,
                            hipMemcpyDeviceToHost, ctx->stream));
+  auto t2_ = tn();
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  if (kvt)
+    fprintf(stderr, "[kv_timing] lookup inner: stage %.2f launch+copyq %.2f sync %.2f ms\n",
+            std::chrono::duration<double, std::milli>(t1_ - t0_).count(),
+            std::chrono::duration<double, std::milli>(t2_ - t1_).count(),
+            std::chrono::duration<double, std::milli>(tn() - t2_).count());
   if (kernel_ms) {
     float ms = 0.f;
     HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
@@ -1408,11 +1419,19 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
   });
   std::vector<uint8_t> entries(n_in_total * 64);
   std::vector<uint64_t> found((n_in_total + 63) / 64, 0);
+  const bool kv_timing = getenv("KV_TIMING") != nullptr;
+  auto tnow = []() { return std::chrono::steady_clock::now(); };
+  auto tms = [](std::chrono::steady_clock::time_point a,
+                std::chrono::steady_clock::time_point b) {
+    return std::chrono::duration<double, std::milli>(b - a).count();
+  };
+  auto t_a = tnow();
   if (n_in_total) {
     int rc = utxo_lookup_nolock(ctx, ops.data(), n_in_total, entries.data(),
                                 found.data(), nullptr);
     if (rc) return rc;
   }
+  auto t_b = tnow();
 
   /* rebuild the blob with populated entries; pre-fail txs with missing inputs */
   std::vector<int32_t> pre_codes(n_txs, 0);
@@ -1491,9 +1510,13 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
     memcpy(dst, blob + outs_start, tx_end - outs_start);
   });
 
+  auto t_c = tnow();
   int rc = validate_block_impl(ctx, pop.data(), pop.size(), pov_daa_score,
                                block_daa_score, flags, tx_codes_out, fees_out,
                                muhash_partial_out, pre_codes.data());
+  if (kv_timing)
+    fprintf(stderr, "[kv_timing] utxo: lookup %.2fms rebuild %.2fms impl %.2fms\n",
+            tms(t_a, t_b), tms(t_b, t_c), tms(t_c, tnow()));
   if (rc || !apply_diff) return rc;
 
   /* diff apply for accepted txs: remove spent, upsert created */
