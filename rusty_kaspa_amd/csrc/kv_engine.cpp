@@ -71,6 +71,10 @@ struct kv_ctx {
    * keyed checks survive across calls — the mempool→block revalidation dedup.
    * Key is a blake2b-256 so a collision is cryptographically excluded. */
   std::unordered_map<kv_sig_key, uint8_t, kv_sig_key_hash> sig_cache;
+  /* reusable populate-path scratch (capacity persists across calls — a fresh
+   * multi-MB vector per block showed up as page-fault spikes in the bench) */
+  std::vector<uint8_t> pop_buf, ops_buf, ent_buf;
+  std::vector<uint64_t> found_buf;
   /* GPU-resident UTXO set */
   kv::utxo_slot *d_utxo = nullptr;
   uint64_t utxo_cap = 0; /* power of two */
@@ -1408,7 +1412,8 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
   for (int t = 0; t < n_txs; t++)
     in_base[t + 1] = in_base[t] + txs[t].inputs.size();
   size_t n_in_total = in_base[n_txs];
-  std::vector<uint8_t> ops(n_in_total * 36);
+  std::vector<uint8_t> &ops = ctx->ops_buf;
+  ops.resize(n_in_total * 36);
   kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
     uint8_t *dst = ops.data() + in_base[t] * 36;
     for (auto &in : txs[t].inputs) {
@@ -1417,8 +1422,10 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
       dst += 36;
     }
   });
-  std::vector<uint8_t> entries(n_in_total * 64);
-  std::vector<uint64_t> found((n_in_total + 63) / 64, 0);
+  std::vector<uint8_t> &entries = ctx->ent_buf;
+  entries.resize(n_in_total * 64);
+  std::vector<uint64_t> &found = ctx->found_buf;
+  found.assign((n_in_total + 63) / 64, 0);
   const bool kv_timing = getenv("KV_TIMING") != nullptr;
   auto tnow = []() { return std::chrono::steady_clock::now(); };
   auto tms = [](std::chrono::steady_clock::time_point a,
@@ -1461,7 +1468,8 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
   size_t hdr_len = 4 + 4ull * n_txs;
   new_off[0] = hdr_len;
   for (int t = 0; t < n_txs; t++) new_off[t + 1] += new_off[t];
-  std::vector<uint8_t> pop(new_off[n_txs]);
+  std::vector<uint8_t> &pop = ctx->pop_buf;
+  pop.resize(new_off[n_txs]);
   memcpy(pop.data(), blob, 4);
   kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
     const HTx &tx = txs[t];
