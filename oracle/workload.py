@@ -159,11 +159,13 @@ def gen_block(oracle, seed: int, n_txs: int, *, pct_multi_input=0, pct_ecdsa=0,
             out_vals.append(v)
             remain -= v
         out_vals.append(remain)
-        outputs = [B.tx_output(v, sg.p2pk_spk(key_seq + 1000 + i))
-                   for i, v in enumerate(out_vals)]
+        out_keys = [key_seq + 1000 + i for i in range(len(out_vals))]
+        outputs = [B.tx_output(v, sg.p2pk_spk(k))
+                   for k, v in zip(out_keys, out_vals)]
         payload = bytes(rng.randrange(256) for _ in range(payload_len))
         txs.append(B.tx_dict(0, inputs, outputs, payload=payload))
-        specs.append({"inputs": in_specs, "invalid": invalid})
+        specs.append({"inputs": in_specs, "invalid": invalid,
+                      "out_keys": out_keys, "out_vals": out_vals})
 
     # pass 1: tx ids (exclude sig scripts, so ids are final before signing)
     blob = B.build_blob(txs)
@@ -198,7 +200,47 @@ def gen_block(oracle, seed: int, n_txs: int, *, pct_multi_input=0, pct_ecdsa=0,
         t["storage_mass"] = storage_mass(t)
 
     blob = B.build_blob(txs)
+    for t in range(n_txs):
+        specs[t]["tx_id"] = txs[t]["tx_id"]
     return blob, {"n_txs": n_txs, "specs": specs}
+
+
+def gen_spend_block(oracle, seed: int, prev_meta, prev_block_daa: int,
+                    accepted=None):
+    """Build a follow-up block whose txs spend the PREVIOUS block's outputs
+    (P2PK schnorr, 1-in/1-out) — the cross-block chain the UTXO-diff apply
+    must sustain (utxo_diff.rs:224). `accepted` filters which prev txs'
+    outputs exist (default: the non-invalid ones). Returns (blob, meta) in the
+    same populated form as gen_block; strip_utxo_entries turns it into the
+    table-resolved shape."""
+    rng = random.Random(seed)
+    sg = Signer(oracle)
+    txs = []
+    chosen = []
+    for t, spec in enumerate(prev_meta["specs"]):
+        ok = spec.get("invalid") is False if accepted is None else accepted[t]
+        if not ok:
+            continue
+        for i, (k, v) in enumerate(zip(spec["out_keys"], spec["out_vals"])):
+            chosen.append((spec["tx_id"], i, k, v))
+    for tx_id, idx, key, value in chosen:
+        spk = sg.p2pk_spk(key)
+        fee = min(1000, value - 1) if value > 1 else 0
+        inp = B.tx_input(tx_id, idx, sequence=0, commit_kind=0, commit_value=1,
+                         utxo=B.utxo_entry(value, spk, prev_block_daa))
+        out = B.tx_output(value - fee, sg.p2pk_spk(key + 500000))
+        txs.append(B.tx_dict(0, [inp], [out]))
+    blob = B.build_blob(txs)
+    for t in range(len(txs)):
+        txs[t]["tx_id"] = sg.tx_id(blob, t)
+    blob = B.build_blob(txs)
+    for t, (tx_id, idx, key, value) in enumerate(chosen):
+        msg = sg.sighash(blob, t, 0)
+        sig = sg.schnorr_sign(key, msg)
+        txs[t]["inputs"][0]["sig_script"] = bytes([0x41]) + sig + bytes([SIGHASH_ALL])
+    for t in txs:
+        t["storage_mass"] = storage_mass(t)
+    return B.build_blob(txs), {"n_txs": len(txs)}
 
 
 STORM = 10**8 * 10**4

@@ -172,3 +172,40 @@ def test_missing_outpoint(oracle, engine):
     if counts[0] > 1:
         found, _ = lookup(engine, [seeds[1][0]])
         assert found[0] == 1
+
+
+def test_chained_blocks_through_table(oracle, engine):
+    """Block B spends block A's created outputs straight from the table
+    (the cross-block UTXO-diff lifecycle, utxo_diff.rs:224), then a replay of
+    B double-spends and every tx fails with MISSING_OUTPOINT."""
+    from workload import gen_spend_block
+    D1, D2 = 10**9, 10**9 + 10
+    nA = 40
+    blobA, metaA = gen_block(oracle, seed=41, n_txs=nA, pct_multi_input=15,
+                             pct_invalid=10)
+    strippedA, seedsA = strip_utxo_entries(blobA)
+    seed_table(engine, seedsA, capacity=4 * len(seedsA))
+    cA, _, _ = engine.validate_block_utxo(strippedA, nA, D1, D1,
+                                          SKIP_MASS, apply_diff=True)
+    icA, _, _ = engine.validate_block(blobA, nA, D1, D1, SKIP_MASS)
+    assert cA == icA
+
+    accepted = [c == 0 for c in cA]
+    blobB, metaB = gen_spend_block(oracle, seed=42, prev_meta=metaA,
+                                   prev_block_daa=D1, accepted=accepted)
+    nB = metaB["n_txs"]
+    assert nB > 10
+    # ground truth: the inline-populated form of B (entries as the diff wrote
+    # them: value/spk from A's outputs, daa = D1, non-coinbase)
+    iB, fB, pB = engine.validate_block(blobB, nB, D2, D2, SKIP_MASS)
+    assert all(c == 0 for c in iB), iB[:5]
+    strippedB, _ = strip_utxo_entries(blobB)
+    uB, ufB, upB = engine.validate_block_utxo(strippedB, nB, D2, D2,
+                                              SKIP_MASS, apply_diff=True)
+    assert uB == iB and ufB == fB and upB == pB
+
+    # replay: every input was just spent
+    rB, rfB, _ = engine.validate_block_utxo(strippedB, nB, D2, D2, SKIP_MASS,
+                                            apply_diff=True, want_muhash=False)
+    assert all(c == MISSING_OUTPOINT for c in rB), rB[:5]
+    assert all(f == 0 for f in rfB)
