@@ -96,18 +96,28 @@ def block_mode(args):
     import ctypes
     sys.path.insert(0, os.path.join(REPO, "oracle"))
     from workload import gen_block
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+    if distributed:
+        import torch
+        import torch.distributed as dist
+        dist.init_process_group(backend="nccl")
+        torch.cuda.set_device(local_rank)
     oracle = load_oracle()
     n_txs = args.block_batch * 300
     t0 = time.time()
-    blob, _ = gen_block(oracle, seed=SEED, n_txs=n_txs, pct_multi_input=20,
+    # each rank validates its own shard of blocks (blocks are independent;
+    # the only cross-rank step in the real path is the muhash combine)
+    blob, _ = gen_block(oracle, seed=SEED + rank, n_txs=n_txs, pct_multi_input=20,
                         pct_ecdsa=10)
     log(f"generated {n_txs} mixed txs ({len(blob)/1e6:.1f} MB blob) "
         f"in {time.time()-t0:.1f}s")
     from rusty_kaspa_amd.engine import Engine
     # sig cache OFF for the bench: the batch is reused across timed steps and
     # a cache hit would skip the verify work being measured
-    eng = Engine(sig_cache_size=0)
-    n_inputs = sum(1 for _ in range(0))  # informational only
+    eng = Engine(device=local_rank, sig_cache_size=0)
 
     utxo = args.mode == "block-utxo"
     if utxo:
@@ -126,26 +136,63 @@ def block_mode(args):
         log(f"UTXO table seeded: {len(seeds)} entries")
         blob = stripped
 
+    if distributed:
+        import torch
+        import torch.distributed as dist
+        gather_buf = [torch.zeros(768, dtype=torch.uint8,
+                                  device=f"cuda:{local_rank}")
+                      for _ in range(world)]
+        mine = torch.zeros(768, dtype=torch.uint8, device=f"cuda:{local_rank}")
+
     def one_step():
         if utxo:
             codes, fees, partial = eng.validate_block_utxo(
                 blob, n_txs, 10**9, 10**9, 0, apply_diff=False)
         else:
             codes, fees, partial = eng.validate_block(blob, n_txs, 10**9, 10**9, 0)
-        mh = eng.muhash_finalize(partial)
+        if distributed:
+            # the real path's one exchange: allgather the 768B muhash
+            # (numerator‖denominator) partials over RCCL, fold the
+            # multiplicative combine on rank 0 (not an RCCL builtin)
+            import torch
+            import torch.distributed as dist
+            mine.copy_(torch.frombuffer(bytearray(partial), dtype=torch.uint8))
+            dist.all_gather(gather_buf, mine)
+            if rank == 0:
+                acc = bytearray(gather_buf[0].cpu().numpy().tobytes())
+                for r in range(1, world):
+                    eng.muhash_combine(acc, gather_buf[r].cpu().numpy().tobytes())
+                mh = eng.muhash_finalize(bytes(acc))
+            else:
+                mh = None
+        else:
+            mh = eng.muhash_finalize(partial)
         return codes, mh
 
     for _ in range(args.warmup):
         codes, _ = one_step()
     assert all(c == 0 for c in codes), "unexpected invalid txs in bench batch"
+    if distributed:
+        import torch
+        import torch.distributed as dist
+        dist.barrier()
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         one_step()
     elapsed = time.perf_counter() - t0
-    value = n_txs * args.steps / elapsed
+    if distributed:
+        import torch
+        import torch.distributed as dist
+        torch.cuda.synchronize()
+        dist.barrier()
+        el = torch.tensor([elapsed], device=f"cuda:{local_rank}")
+        dist.all_reduce(el, op=dist.ReduceOp.MAX)
+        elapsed = float(el.item())
+    value = world * n_txs * args.steps / elapsed
 
     cpu_baseline = None
-    if not args.skip_cpu_baseline:
+    if not args.skip_cpu_baseline and world == 1:
         cores = os.cpu_count() or 8
         codes_a = (ctypes.c_int32 * n_txs)()
         fees_a = (ctypes.c_uint64 * n_txs)()
@@ -162,7 +209,7 @@ def block_mode(args):
         "metric": "txs-validated/sec",
         "value": round(value, 1),
         "unit": "txs/s",
-        "n_gpus": 1,
+        "n_gpus": world,
         "steps": args.steps,
         "warmup": args.warmup,
         "ms_per_step": round(elapsed / args.steps * 1000, 3),
@@ -175,12 +222,16 @@ def block_mode(args):
                                 else "block-validate-config3"),
                    "blocks_per_step": args.block_batch, "txs_per_block": 300,
                    "mix": "70p2pk/20multi-in/10ecdsa", "flags": "FULL", "sig_cache": "off",
-                   "parallelism": "single"},
+                   "parallelism": f"dp{world}" if world > 1 else "single"},
         "roofline": None,  # per-kernel rooflines live in the default mode + profiles/
         "cpu_baseline": cpu_baseline,
     }
-    print(json.dumps(result), flush=True)
+    if rank == 0:
+        print(json.dumps(result), flush=True)
     eng.close()
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
 
 
 def main():

@@ -81,3 +81,4 @@ def test_bitmap_and_muhash_exchange(oracle):
     exp = (ctypes.c_uint8 * 32)()
     oracle.ok_muhash_finalize(num, den, exp)
     assert m0 == bytes(exp)
+
