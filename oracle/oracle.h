@@ -149,3 +149,15 @@ int ok_muhash_add_tx(const uint8_t *blob, size_t blob_len, uint32_t tx_index,
 #endif
 
 #endif /* OK_ORACLE_H */
+
+/* ---------------- Merkle root + body-in-isolation (oracle) ---------------- */
+#define OK_BODY_DUP_TX 10
+#define OK_BODY_DOUBLE_SPEND 11
+#define OK_BODY_CHAINED 12
+/* calc_merkle_root (crypto/merkle/src/lib.rs:13-52) over 32B leaves */
+void ok_merkle_root(const uint8_t *hashes, size_t n, uint8_t out32[32]);
+/* calc_hash_merkle_root over a blob's txs (consensus/core/src/merkle.rs:5) */
+int ok_blob_merkle_root(const uint8_t *blob, size_t blob_len, uint8_t out32[32]);
+/* duplicate-tx / in-block double-spend / chained-tx checks
+ * (body_validation_in_isolation.rs:126-173); 0 ok or OK_BODY_* */
+int ok_body_check(const uint8_t *blob, size_t blob_len);
