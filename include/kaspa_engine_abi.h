@@ -83,6 +83,11 @@ extern "C" {
 #define KV_ERR_MASS_INCOMPUTABLE 8
 #define KV_ERR_MISSING_OUTPOINT 9   /* populate failure: outpoint not in the UTXO set
                                        (ruleerrors::RuleError::MissingTxOutpoints) */
+/* body-in-isolation rule codes (RuleError::{DuplicateTransactions,
+ * DoubleSpendInSameBlock, ChainedTransaction}) */
+#define KV_ERR_BODY_DUP_TX 10
+#define KV_ERR_BODY_DOUBLE_SPEND 11
+#define KV_ERR_BODY_CHAINED 12
 #define KV_ERR_BAD_BLOB 90
 #define KV_ERR_SIGNATURE_INVALID_BASE 100
 #define KV_ERR_SIGNATURE_EMPTY_BASE 200
@@ -228,6 +233,14 @@ int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
                            uint64_t pov_daa_score, uint64_t block_daa_score,
                            uint32_t flags, int apply_diff, int32_t *tx_codes_out,
                            uint64_t *fees_out, uint8_t *muhash_partial_out);
+
+/* Block-body-in-isolation batch (body_validation_in_isolation.rs):
+ * merkle_root_out ← calc_hash_merkle_root over the txs (GPU leaf hashes,
+ * caller compares against the header commitment); rule_code_out ← 0 or the
+ * first KV_ERR_BODY_* violation (duplicate tx / in-block double spend /
+ * chained tx). Either out-pointer may be NULL. */
+int kv_block_body_check(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
+                        uint8_t merkle_root_out[32], int32_t *rule_code_out);
 
 /* Sig-cache statistics (crypto/txscript/src/caches.rs:57-82 counters). */
 typedef struct {

@@ -496,7 +496,7 @@ struct DeviceBuf {
 
 struct ValidateBufs {
   DeviceBuf blob, subhashes, s_jobs, e_jobs, s_tuples, e_tuples, s_bitmap, e_bitmap,
-      s_status, e_status, elem_jobs, elements, partials_a, partials_b;
+      s_status, e_status, elem_jobs, elements, partials_a, partials_b, tx_hashes;
 };
 
 static ValidateBufs g_vb; /* guarded by ctx->mu (single validate at a time) */
@@ -1371,6 +1371,136 @@ extern "C" int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
   std::lock_guard<std::mutex> lk(ctx->mu);
   return utxo_lookup_nolock(ctx, outpoints, n, entries_out, found_bitmap,
                             kernel_ms);
+}
+
+
+/* ---------------- merkle root + body-in-isolation batch ----------------
+ * ⇔ validate_body_in_isolation (consensus/src/pipeline/body_processor/
+ * body_validation_in_isolation.rs): calc_hash_merkle_root (:38 via
+ * consensus/core/src/merkle.rs:5 → crypto/merkle/src/lib.rs:31-52) over the
+ * GPU-computed tx hashes, plus check_duplicate_transactions (:152),
+ * check_block_double_spends (:126) and check_no_chained_transactions (:136).
+ * The leaf hashes (one keyed blake2b per tx over the full serialization) run
+ * on device; the log-depth fold and the set checks run on the host pool. */
+extern "C" int kv_block_body_check(kv_ctx *ctx, const uint8_t *blob,
+                                   size_t blob_len, uint8_t merkle_root_out[32],
+                                   int32_t *rule_code_out) {
+  if (!ctx) {
+    set_error("kv_block_body_check: null ctx");
+    return -1;
+  }
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  vector<HTx> txs;
+  int n_txs = parse_blob_host(blob, blob_len, txs);
+  if (n_txs < 0) {
+    set_error("kv_block_body_check: malformed blob");
+    return -1;
+  }
+  /* leaf hashes on device */
+  std::vector<uint8_t> hashes((size_t)n_txs * 32);
+  if (n_txs > 0) {
+    if (g_vb.blob.ensure(blob_len) || g_vb.tx_hashes.ensure((size_t)n_txs * 32))
+      return -2;
+    HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
+                             ctx->stream));
+    hipLaunchKernelGGL(kv::kv_tx_hash_kernel, dim3((n_txs + 255) / 256),
+                       dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.blob.p,
+                       (uint32_t)n_txs, (uint8_t *)g_vb.tx_hashes.p);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipMemcpyAsync(hashes.data(), g_vb.tx_hashes.p,
+                             (size_t)n_txs * 32, hipMemcpyDeviceToHost,
+                             ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  }
+  /* merkle fold (crypto/merkle/src/lib.rs:31-52): pad to a power of two,
+   * absent left → absent parent, absent right → ZERO_HASH */
+  if (merkle_root_out) {
+    if (n_txs == 0) {
+      memset(merkle_root_out, 0, 32);
+    } else if (n_txs == 1) {
+      memcpy(merkle_root_out, hashes.data(), 32);
+    } else {
+      size_t pot = 1;
+      while (pot < (size_t)n_txs) pot <<= 1;
+      std::vector<uint8_t> cur(pot * 32, 0), pres(pot, 0);
+      memcpy(cur.data(), hashes.data(), (size_t)n_txs * 32);
+      for (int i = 0; i < n_txs; i++) pres[i] = 1;
+      static const uint8_t ZERO[32] = {0};
+      size_t width = pot;
+      while (width > 1) {
+        kvh_parallel_for((uint32_t)(width / 2), [&](uint32_t o) {
+          size_t i = 2 * (size_t)o;
+          if (!pres[i]) {
+            pres[o] = 0;
+            return;
+          }
+          uint8_t buf[64];
+          memcpy(buf, cur.data() + i * 32, 32);
+          memcpy(buf + 32, pres[i + 1] ? cur.data() + (i + 1) * 32 : ZERO, 32);
+          uint8_t h[32];
+          static const uint8_t MKEY[] = "MerkleBranchHash";
+          h_blake2b_keyed(MKEY, sizeof(MKEY) - 1, buf, 64, h);
+          memcpy(cur.data() + o * 32, h, 32);
+          pres[o] = 1;
+        });
+        width /= 2;
+      }
+      memcpy(merkle_root_out, cur.data(), 32);
+    }
+  }
+  /* body rule checks (first violation wins, reference order) */
+  int32_t code = 0;
+  {
+    std::vector<std::array<uint8_t, 32>> ids(n_txs);
+    for (int t = 0; t < n_txs; t++) memcpy(ids[t].data(), txs[t].tx_id, 32);
+    std::sort(ids.begin(), ids.end());
+    for (int t = 0; t + 1 < n_txs; t++)
+      if (ids[t] == ids[t + 1]) {
+        code = KV_ERR_BODY_DUP_TX;
+        break;
+      }
+  }
+  if (!code) {
+    std::vector<std::array<uint8_t, 36>> ops;
+    for (auto &tx : txs)
+      for (auto &in : tx.inputs) {
+        std::array<uint8_t, 36> o;
+        memcpy(o.data(), in.prev_tx_id, 32);
+        memcpy(o.data() + 32, &in.prev_index, 4);
+        ops.push_back(o);
+      }
+    std::sort(ops.begin(), ops.end());
+    for (size_t i = 0; i + 1 < ops.size(); i++)
+      if (ops[i] == ops[i + 1]) {
+        code = KV_ERR_BODY_DOUBLE_SPEND;
+        break;
+      }
+    if (!code) {
+      std::vector<std::array<uint8_t, 36>> created;
+      for (auto &tx : txs)
+        for (uint32_t i = 0; i < tx.outputs.size(); i++) {
+          std::array<uint8_t, 36> o;
+          memcpy(o.data(), tx.tx_id, 32);
+          memcpy(o.data() + 32, &i, 4);
+          created.push_back(o);
+        }
+      std::sort(created.begin(), created.end());
+      for (auto &tx : txs) {
+        for (auto &in : tx.inputs) {
+          std::array<uint8_t, 36> key;
+          memcpy(key.data(), in.prev_tx_id, 32);
+          memcpy(key.data() + 32, &in.prev_index, 4);
+          if (std::binary_search(created.begin(), created.end(), key)) {
+            code = KV_ERR_BODY_CHAINED;
+            break;
+          }
+        }
+        if (code) break;
+      }
+    }
+  }
+  if (rule_code_out) *rule_code_out = code;
+  return 0;
 }
 
 /* ---------------- populate + validate + diff-apply ----------------

@@ -328,6 +328,72 @@ extern "C" __global__ void kv_u3072_reduce_kernel(const uint64_t *__restrict__ e
   for (int k = 0; k < KVU_LIMBS; k++) partials[(size_t)t * KVU_LIMBS + k] = acc.l[k];
 }
 
+
+/* hashing::tx::hash ⇔ consensus/core/src/hashing/tx.rs:20-24 (crescendo):
+ * keyed blake2b-256("TransactionHash") over write_transaction with signature
+ * scripts, per-version commit fields, payload var-bytes and the mass rule
+ * (v0: mass hashed only when > 0; v1+: always) — one tx per lane. */
+extern "C" __global__ void kv_tx_hash_kernel(const uint8_t *__restrict__ blob,
+                                             uint32_t n_txs,
+                                             uint8_t *__restrict__ hashes_out) {
+  uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= n_txs) return;
+  blob_tx tx;
+  blob_tx_at(blob, t, tx);
+  uint64_t storage_mass = bl_rd64(tx.base + 48);
+  static const uint8_t KEY[] = "TransactionHash";
+  b2b_state S;
+  b2b_init_keyed(S, KEY, sizeof(KEY) - 1);
+  b2b_update_u16(S, tx.version);
+  b2b_update_u64(S, tx.n_inputs);
+  const uint8_t *p = tx.inputs0;
+  for (uint32_t i = 0; i < tx.n_inputs; i++) {
+    blob_input in;
+    blob_input_at(p, in);
+    b2b_update(S, in.prev_tx_id, 32);
+    b2b_update_u32(S, in.prev_index);
+    b2b_update_u64(S, in.sig_script_len);
+    b2b_update(S, in.sig_script, in.sig_script_len);
+    if (tx.version < 1) {
+      uint8_t soc = in.commit_kind == 0 ? (uint8_t)in.commit_value : 0;
+      b2b_update(S, &soc, 1);
+    }
+    b2b_update_u64(S, in.sequence);
+    if (tx.version >= 1)
+      b2b_update_u16(S, in.commit_kind == 1 ? in.commit_value : 0);
+    p = in.end;
+  }
+  b2b_update_u64(S, tx.n_outputs);
+  for (uint32_t i = 0; i < tx.n_outputs; i++) {
+    blob_output o;
+    blob_output_at(p, o);
+    b2b_update_u64(S, o.value);
+    b2b_update_u16(S, o.spk_version);
+    b2b_update_u64(S, o.spk_len);
+    b2b_update(S, o.spk, o.spk_len);
+    if (tx.version >= 1) {
+      uint8_t hc = o.has_covenant ? 1 : 0;
+      b2b_update(S, &hc, 1);
+      if (o.has_covenant) {
+        b2b_update_u16(S, o.cov_auth_input);
+        b2b_update(S, o.cov_id, 32);
+      }
+    }
+    p = o.end;
+  }
+  b2b_update_u64(S, tx.lock_time);
+  b2b_update(S, tx.subnetwork_id, 20);
+  b2b_update_u64(S, tx.gas);
+  b2b_update_u64(S, tx.payload_len);
+  b2b_update(S, tx.payload, tx.payload_len);
+  if (tx.version < 1) {
+    if (storage_mass > 0) b2b_update_u64(S, storage_mass);
+  } else {
+    b2b_update_u64(S, storage_mass);
+  }
+  b2b_final(S, hashes_out + (size_t)t * 32);
+}
+
 } // namespace kv
 
 /* ---------------- wave-cooperative U3072 mulmod ----------------

@@ -139,3 +139,13 @@ class Engine:
         self._check(self.lib.kv_sig_cache_stats(ctypes.c_void_p(self.ctx),
                                                 ctypes.byref(out)))
         return out.insertions, out.hits, out.misses
+
+    def block_body_check(self, blob: bytes):
+        """Merkle root over tx hashes + duplicate/double-spend/chained checks."""
+        root = (ctypes.c_uint8 * 32)()
+        code = ctypes.c_int32()
+        rc = self.lib.kv_block_body_check(
+            ctypes.c_void_p(self.ctx), blob, ctypes.c_size_t(len(blob)),
+            root, ctypes.byref(code))
+        self._check(rc)
+        return bytes(root), code.value
