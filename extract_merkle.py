@@ -67,10 +67,24 @@ def build(mass0):
         out.append(d)
     return B.build_blob(out)
 
-blob0 = build(0)
-blob7 = build(7)
 import ctypes
 O = ctypes.CDLL('/root/repo/oracle/liboracle.so')
+
+def with_ids(blob_builder):
+    """fill the carried tx_id fields (the blob contract: id is carried)"""
+    import struct as _s
+    blob = blob_builder()
+    n, = _s.unpack_from("<I", blob, 0)
+    offs = list(_s.unpack_from(f"<{n}I", blob, 4))
+    out = bytearray(blob)
+    for t in range(n):
+        idb = (ctypes.c_uint8 * 32)()
+        assert O.ok_tx_id(bytes(blob), len(blob), t, idb) == 0
+        out[offs[t] + 56:offs[t] + 88] = bytes(idb)
+    return bytes(out)
+
+blob0 = with_ids(lambda: build(0))
+blob7 = with_ids(lambda: build(7))
 got = (ctypes.c_uint8 * 32)()
 assert O.ok_blob_merkle_root(blob0, len(blob0), got) == 0
 print("root(mass0):", bytes(got).hex())
