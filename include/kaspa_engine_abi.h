@@ -234,6 +234,18 @@ int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
                            uint32_t flags, int apply_diff, int32_t *tx_codes_out,
                            uint64_t *fees_out, uint8_t *muhash_partial_out);
 
+/* Mempool batch validation ⇔ validate_mempool_transaction_in_utxo_context
+ * (utxo_validation.rs:418-457) fanned over independent txs: SkipMassCheck +
+ * per-tx COMPUTED contextual storage mass + optional feerate threshold
+ * (<= 0 disables; fee / normalized_max(mass) <= threshold →
+ * KV_ERR_FEERATE_TOO_LOW). from_utxo_table = 1 resolves entries from the
+ * GPU-resident table (missing outpoint → KV_ERR_MISSING_OUTPOINT), 0 takes
+ * them inline from the blob. */
+int kv_validate_mempool(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
+                        uint64_t pov_daa_score, double feerate_threshold,
+                        int from_utxo_table, int32_t *tx_codes_out,
+                        uint64_t *fees_out);
+
 /* Block-body-in-isolation batch (body_validation_in_isolation.rs):
  * merkle_root_out ← calc_hash_merkle_root over the txs (GPU leaf hashes,
  * caller compares against the header commitment); rule_code_out ← 0 or the

@@ -354,3 +354,88 @@ int ok_gen_ecdsa_tuples(uint64_t seed, size_t n, uint32_t invalid_permille,
   }
   return 0;
 }
+
+/* ---- mempool validation ⇔ validate_mempool_transaction_in_utxo_context
+ * (utxo_validation.rs:418-457): contextual storage mass is COMPUTED (the
+ * carried commitment is ignored — the mempool sets it), validation runs with
+ * SkipMassCheck, and the optional feerate threshold compares
+ * fee / normalized_max(mass) <= threshold → FeerateTooLow
+ * (tx_validation_in_utxo_context.rs:69-77). Normalization uses the mainnet
+ * cofactors (mass/mod.rs:177-190,298-308; params.rs:623-626: limits
+ * {compute 500k, storage 500k, transient 1M} → storage cofactor 1.0,
+ * transient 0.5), mass_per_tx_byte 1, mass_per_script_pub_key_byte 10,
+ * GRAMS_PER_SIGOP_COUNT_UNIT 1000, GRAMS_PER_COMPUTE_BUDGET_UNIT 100,
+ * TRANSIENT_BYTE_TO_MASS_FACTOR 4 (constants.rs:31, mass/units.rs:4-5). */
+
+static uint64_t mp_estimated_size(const ok_tx *tx) {
+  /* transaction_estimated_serialized_size (mass/mod.rs:21-78) */
+  uint64_t size = 2 + 8 + 8 + 8 + 20 + 8 + 32 + 8 + tx->payload_len;
+  for (uint32_t i = 0; i < tx->n_inputs; i++) {
+    size += 32 + 4 + 8 + tx->inputs[i].sig_script_len + 8;
+    if (tx->version >= 1) size += 2;
+  }
+  for (uint32_t i = 0; i < tx->n_outputs; i++) {
+    size += 8 + 2 + 8 + tx->outputs[i].spk_len;
+    if (tx->outputs[i].has_covenant) size += 2 + 32;
+  }
+  return size;
+}
+
+static uint64_t mp_normalized_mass(const ok_tx *tx, uint64_t storage_mass) {
+  if (ok_tx_is_coinbase(tx)) return storage_mass; /* non-contextual = 0 */
+  uint64_t size = mp_estimated_size(tx);
+  uint64_t spk_bytes = 0;
+  for (uint32_t i = 0; i < tx->n_outputs; i++)
+    spk_bytes += 2 + tx->outputs[i].spk_len;
+  uint64_t script_mass = 0;
+  if (tx->version >= 1) {
+    for (uint32_t i = 0; i < tx->n_inputs; i++)
+      script_mass += 100ull * (tx->inputs[i].commit_kind == 1
+                                   ? tx->inputs[i].commit_value
+                                   : 0);
+  } else {
+    for (uint32_t i = 0; i < tx->n_inputs; i++)
+      script_mass += 1000ull * (tx->inputs[i].commit_kind == 0
+                                    ? tx->inputs[i].commit_value
+                                    : 0);
+  }
+  uint64_t compute = size * 1 + spk_bytes * 10 + script_mass;
+  uint64_t transient = size * 4;
+  /* normalized_max: max(ceil(storage*1.0), compute, ceil(transient*0.5)) */
+  uint64_t tnorm = (transient + 1) / 2;
+  uint64_t m = storage_mass;
+  if (compute > m) m = compute;
+  if (tnorm > m) m = tnorm;
+  return m;
+}
+
+int ok_validate_mempool(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_score,
+                        double feerate_threshold, int threads,
+                        int32_t *tx_codes_out, uint64_t *fees_out) {
+  uint8_t mh[32];
+  int rc = ok_validate_block_parallel(blob, blob_len, pov_daa_score, pov_daa_score,
+                                      KV_FLAGS_SKIP_MASS_CHECK, threads,
+                                      tx_codes_out, fees_out, mh);
+  if (rc != 0) return rc;
+  uint32_t n_txs;
+  memcpy(&n_txs, blob, 4);
+  for (uint32_t t = 0; t < n_txs; t++) {
+    ok_tx tx;
+    if (ok_tx_parse(blob, blob_len, t, &tx) != 0) return -1;
+    uint64_t storage = 0;
+    if (calc_storage_mass_tx(&tx, &storage)) {
+      /* contextual mass computed BEFORE validation in the reference —
+       * MassIncomputable wins over any validation error */
+      tx_codes_out[t] = KV_ERR_MASS_INCOMPUTABLE;
+      fees_out[t] = 0;
+    } else if (tx_codes_out[t] == 0 && feerate_threshold > 0) {
+      uint64_t m = mp_normalized_mass(&tx, storage);
+      if (m > 0 && (double)fees_out[t] / (double)m <= feerate_threshold) {
+        tx_codes_out[t] = KV_ERR_FEERATE_TOO_LOW;
+        fees_out[t] = 0;
+      }
+    }
+    ok_tx_free(&tx);
+  }
+  return 0;
+}

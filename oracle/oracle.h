@@ -161,3 +161,11 @@ int ok_blob_merkle_root(const uint8_t *blob, size_t blob_len, uint8_t out32[32])
 /* duplicate-tx / in-block double-spend / chained-tx checks
  * (body_validation_in_isolation.rs:126-173); 0 ok or OK_BODY_* */
 int ok_body_check(const uint8_t *blob, size_t blob_len);
+
+/* mempool batch validation ⇔ validate_mempool_transaction_in_utxo_context
+ * (utxo_validation.rs:418-457): SkipMassCheck + computed contextual mass +
+ * optional feerate threshold (<=0 disables; fee/normalized_mass <= threshold
+ * → KV_ERR_FEERATE_TOO_LOW). */
+int ok_validate_mempool(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_score,
+                        double feerate_threshold, int threads,
+                        int32_t *tx_codes_out, uint64_t *fees_out);

@@ -1374,6 +1374,81 @@ extern "C" int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
 }
 
 
+
+static int populate_blob_nolock(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
+                                std::vector<kvhost::HTx> &txs, int n_txs,
+                                std::vector<int32_t> &pre_codes);
+
+/* ---------------- mempool batch validation ----------------
+ * ⇔ validate_mempool_transaction_in_utxo_context (utxo_validation.rs:418-457)
+ * fanned over a batch: entries resolve inline from the blob (from_utxo_table
+ * = 0, the caller populated) or from the GPU-resident table (= 1; a missing
+ * outpoint fails that tx with KV_ERR_MISSING_OUTPOINT, mirroring
+ * populate_mempool_transaction_in_utxo_context:392-415). The contextual
+ * storage mass is COMPUTED per tx (the carried commitment is ignored — the
+ * mempool sets it), validation runs with SkipMassCheck, and the optional
+ * feerate threshold rejects fee / normalized_max(mass) <= threshold with
+ * KV_ERR_FEERATE_TOO_LOW (tx_validation_in_utxo_context.rs:69-77). */
+extern "C" int kv_validate_mempool(kv_ctx *ctx, const uint8_t *blob,
+                                   size_t blob_len, uint64_t pov_daa_score,
+                                   double feerate_threshold, int from_utxo_table,
+                                   int32_t *tx_codes_out, uint64_t *fees_out) {
+  if (!ctx) {
+    set_error("kv_validate_mempool: null ctx");
+    return -1;
+  }
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  vector<HTx> txs;
+  int n_txs = parse_blob_host(blob, blob_len, txs);
+  if (n_txs < 0) {
+    set_error("kv_validate_mempool: malformed blob");
+    return -1;
+  }
+  std::vector<int32_t> pre_codes(n_txs, 0);
+  const uint8_t *vblob = blob;
+  size_t vlen = blob_len;
+  vector<HTx> ptxs;
+  const vector<HTx> *use_txs = &txs;
+  if (from_utxo_table) {
+    if (!ctx->d_utxo) {
+      set_error("kv_validate_mempool: call kv_utxo_reset first");
+      return -1;
+    }
+    int rc = populate_blob_nolock(ctx, blob, blob_len, txs, n_txs, pre_codes);
+    if (rc) return rc;
+    vblob = ctx->pop_buf.data();
+    vlen = ctx->pop_buf.size();
+    if (parse_blob_host(vblob, vlen, ptxs) != n_txs) {
+      set_error("kv_validate_mempool: internal rebuild parse");
+      return -1;
+    }
+    use_txs = &ptxs;
+  }
+  int rc = validate_block_impl(ctx, vblob, vlen, pov_daa_score, pov_daa_score,
+                               KV_FLAGS_SKIP_MASS_CHECK, tx_codes_out, fees_out,
+                               nullptr, pre_codes.data());
+  if (rc) return rc;
+  /* contextual mass + feerate post-pass (mass computed BEFORE validation in
+   * the reference, so MassIncomputable wins over any validation error) */
+  for (int t = 0; t < n_txs; t++) {
+    const HTx &tx = (*use_txs)[t];
+    if (pre_codes[t]) continue; /* populate failure stands */
+    uint64_t storage = 0;
+    if (kvh_storage_mass(tx, h_is_coinbase(tx), &storage)) {
+      tx_codes_out[t] = KV_ERR_MASS_INCOMPUTABLE;
+      fees_out[t] = 0;
+    } else if (tx_codes_out[t] == 0 && feerate_threshold > 0) {
+      uint64_t m = kvh_normalized_mass(tx, storage);
+      if (m > 0 &&
+          (double)fees_out[t] / (double)m <= feerate_threshold) {
+        tx_codes_out[t] = KV_ERR_FEERATE_TOO_LOW;
+        fees_out[t] = 0;
+      }
+    }
+  }
+  return 0;
+}
+
 /* ---------------- merkle root + body-in-isolation batch ----------------
  * ⇔ validate_body_in_isolation (consensus/src/pipeline/body_processor/
  * body_validation_in_isolation.rs): calc_hash_merkle_root (:38 via
@@ -1514,27 +1589,12 @@ extern "C" int kv_block_body_check(kv_ctx *ctx, const uint8_t *blob,
  * GPU-resident table (kv_utxo_reset/upsert); the engine rebuilds a populated
  * blob internally (one linear pass — the entry fields are the only part that
  * moves) and runs the standard pipeline on it. */
-extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
-                                      size_t blob_len, uint64_t pov_daa_score,
-                                      uint64_t block_daa_score, uint32_t flags,
-                                      int apply_diff, int32_t *tx_codes_out,
-                                      uint64_t *fees_out,
-                                      uint8_t *muhash_partial_out) {
-  if (!ctx) {
-    set_error("kv_validate_block_utxo: null ctx");
-    return -1;
-  }
-  std::lock_guard<std::mutex> lk(ctx->mu);
-  if (!ctx->d_utxo) {
-    set_error("kv_validate_block_utxo: call kv_utxo_reset first");
-    return -1;
-  }
-  vector<HTx> txs;
-  int n_txs = parse_blob_host(blob, blob_len, txs);
-  if (n_txs < 0) {
-    set_error("kv_validate_block_utxo: malformed blob");
-    return -1;
-  }
+/* caller holds ctx->mu. Resolves every input's entry from the GPU table into
+ * a rebuilt populated blob (ctx->pop_buf); missing outpoints pre-fail their tx
+ * with KV_ERR_MISSING_OUTPOINT in pre_codes. */
+static int populate_blob_nolock(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
+                                vector<HTx> &txs, int n_txs,
+                                std::vector<int32_t> &pre_codes) {
 
   /* populate: one GPU lookup over every input's outpoint. The gather and the
    * blob rebuild fan over the host pool (per-tx prefix-summed offsets). */
@@ -1571,7 +1631,6 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
   auto t_b = tnow();
 
   /* rebuild the blob with populated entries; pre-fail txs with missing inputs */
-  std::vector<int32_t> pre_codes(n_txs, 0);
   std::vector<size_t> new_off(n_txs + 1, 0);
   kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
     const HTx &tx = txs[t];
@@ -1648,13 +1707,40 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
     memcpy(dst, blob + outs_start, tx_end - outs_start);
   });
 
-  auto t_c = tnow();
-  int rc = validate_block_impl(ctx, pop.data(), pop.size(), pov_daa_score,
-                               block_daa_score, flags, tx_codes_out, fees_out,
-                               muhash_partial_out, pre_codes.data());
   if (kv_timing)
-    fprintf(stderr, "[kv_timing] utxo: lookup %.2fms rebuild %.2fms impl %.2fms\n",
-            tms(t_a, t_b), tms(t_b, t_c), tms(t_c, tnow()));
+    fprintf(stderr, "[kv_timing] utxo populate: lookup %.2fms rebuild %.2fms\n",
+            tms(t_a, t_b), tms(t_b, tnow()));
+  return 0;
+}
+
+extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
+                                      size_t blob_len, uint64_t pov_daa_score,
+                                      uint64_t block_daa_score, uint32_t flags,
+                                      int apply_diff, int32_t *tx_codes_out,
+                                      uint64_t *fees_out,
+                                      uint8_t *muhash_partial_out) {
+  if (!ctx) {
+    set_error("kv_validate_block_utxo: null ctx");
+    return -1;
+  }
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  if (!ctx->d_utxo) {
+    set_error("kv_validate_block_utxo: call kv_utxo_reset first");
+    return -1;
+  }
+  vector<HTx> txs;
+  int n_txs = parse_blob_host(blob, blob_len, txs);
+  if (n_txs < 0) {
+    set_error("kv_validate_block_utxo: malformed blob");
+    return -1;
+  }
+  std::vector<int32_t> pre_codes(n_txs, 0);
+  int rc = populate_blob_nolock(ctx, blob, blob_len, txs, n_txs, pre_codes);
+  if (rc) return rc;
+  std::vector<uint8_t> &pop = ctx->pop_buf;
+  rc = validate_block_impl(ctx, pop.data(), pop.size(), pov_daa_score,
+                           block_daa_score, flags, tx_codes_out, fees_out,
+                           muhash_partial_out, pre_codes.data());
   if (rc || !apply_diff) return rc;
 
   /* diff apply for accepted txs: remove spent, upsert created */
@@ -1662,10 +1748,12 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
   for (int t = 0; t < n_txs; t++) {
     const HTx &tx = txs[t];
     if (tx_codes_out[t] != 0) continue;
-    size_t in_idx = in_base[t];
-    for (size_t i = 0; i < tx.inputs.size(); i++, in_idx++)
-      del_ops.insert(del_ops.end(), ops.begin() + in_idx * 36,
-                     ops.begin() + (in_idx + 1) * 36);
+    for (auto &in : tx.inputs) {
+      uint8_t op[36];
+      memcpy(op, in.prev_tx_id, 32);
+      memcpy(op + 32, &in.prev_index, 4);
+      del_ops.insert(del_ops.end(), op, op + 36);
+    }
     for (uint32_t i = 0; i < tx.outputs.size(); i++) {
       const HOutput &o = tx.outputs[i];
       if (o.spk_len > 36) {

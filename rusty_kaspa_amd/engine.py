@@ -149,3 +149,15 @@ class Engine:
             root, ctypes.byref(code))
         self._check(rc)
         return bytes(root), code.value
+
+    def validate_mempool(self, blob: bytes, n_txs: int, pov_daa: int,
+                         feerate_threshold: float = 0.0,
+                         from_utxo_table: bool = False):
+        codes = (ctypes.c_int32 * n_txs)()
+        fees = (ctypes.c_uint64 * n_txs)()
+        rc = self.lib.kv_validate_mempool(
+            ctypes.c_void_p(self.ctx), blob, ctypes.c_size_t(len(blob)),
+            ctypes.c_uint64(pov_daa), ctypes.c_double(feerate_threshold),
+            ctypes.c_int(1 if from_utxo_table else 0), codes, fees)
+        self._check(rc)
+        return list(codes), list(fees)
