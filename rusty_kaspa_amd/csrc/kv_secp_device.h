@@ -159,7 +159,11 @@ __device__ __forceinline__ void fe_mul(fe &r, const fe &a, const fe &b) {
 #define KV26_R0 ((u64)0x3D10) /* 2^260 mod p = R0 + 2^36 (C<<4) */
 
 struct fe26 {
+#ifdef KV_FE26_ALIGN
+  alignas(16) u32 l[10];
+#else
   u32 l[10];
+#endif
 };
 
 /* mul/sqr safety bound: limbs <= 2*8*2^26 = 2^30 ("magnitude 8" with the

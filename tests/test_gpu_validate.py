@@ -98,3 +98,15 @@ def test_empty_and_edge_blobs(oracle, engine):
     mh = engine.muhash_finalize(partial)
     assert mh.hex() == (
         "544eb3142c000f0ad2c76ac41f4222abbababed830eeafee4b6dc56b52d5cac0")
+
+
+def test_malformed_blob_fails_cleanly(oracle, engine):
+    """Truncated/corrupt blobs error out (<0 rc → RuntimeError) without
+    wedging the context; the next valid call succeeds."""
+    import pytest as _pytest
+    blob, _ = gen_block(oracle, seed=15, n_txs=8)
+    for cut in (0, 3, 10, len(blob) // 2, len(blob) - 1):
+        with _pytest.raises(RuntimeError):
+            engine.validate_block(blob[:cut], 8, 10**9, 10**9, SKIP_MASS)
+    codes, _, _ = engine.validate_block(blob, 8, 10**9, 10**9, SKIP_MASS)
+    assert all(c == 0 for c in codes)
