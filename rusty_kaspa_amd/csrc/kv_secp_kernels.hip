@@ -249,8 +249,8 @@ __device__ __forceinline__ u64 glv_digit(const glv_half &h, int w) {
 /* R = gs·G + ps·P via GLV-split 4-bit windows: 33 window steps of 4 doublings
  * + 4 selected adds (G, φG, P, φP streams; φ applied at add time as one β·x
  * field multiply; negative half-scalars negate the added point's y). */
-__device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const ge &P,
-                                     volatile int *progress = nullptr) {
+__device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
+                                     const ge &P) {
   glv_half g1h, g2h, p1h, p2h;
   glv_split(gs, g1h, g2h);
   glv_split(ps, p1h, p2h);
@@ -270,7 +270,6 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps, const g
     fe26_from_fe(beta, bu);
   }
   gej_set_infinity(R);
-  if (progress) *progress = 800;
 #pragma unroll 1
   for (int w = 32; w >= 0; w--) {
     gej t;
