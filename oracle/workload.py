@@ -93,7 +93,8 @@ class Signer:
 
 
 def gen_block(oracle, seed: int, n_txs: int, *, pct_multi_input=0, pct_ecdsa=0,
-              pct_invalid=0, pct_multisig=0, payload_len=0, utxo_daa=1000):
+              pct_invalid=0, pct_multisig=0, payload_len=0, utxo_daa=1000,
+              pct_alt_hashtype=0):
     """Build a block's worth of independent txs + populated entries.
 
     Returns (blob_bytes, meta) where meta notes which txs carry invalid sigs.
@@ -146,7 +147,13 @@ def gen_block(oracle, seed: int, n_txs: int, *, pct_multi_input=0, pct_ecdsa=0,
                                          commit_value=n,
                                          utxo=B.utxo_entry(amount, spk, utxo_daa)))
                 continue
-            in_specs.append({"kind": kind, "key": key})
+            htype = SIGHASH_ALL
+            if rng.randrange(100) < pct_alt_hashtype:
+                # the other five valid SigHashTypes (sighash_type.rs:15-22 —
+                # bit flags, so Single is 0x04): None, Single, and the
+                # AnyOneCanPay variants
+                htype = rng.choice([0x02, 0x04, 0x81, 0x82, 0x84])
+            in_specs.append({"kind": kind, "key": key, "htype": htype})
             inputs.append(B.tx_input(prev, rng.randrange(4), sequence=0,
                                      commit_kind=0, commit_value=1,
                                      utxo=B.utxo_entry(amount, spk, utxo_daa)))
@@ -185,9 +192,10 @@ def gen_block(oracle, seed: int, n_txs: int, *, pct_multi_input=0, pct_ecdsa=0,
                 sig_script += bytes([0x4C, len(ins["redeem"])]) + ins["redeem"]
             else:
                 ecdsa = ins["kind"] == "ecdsa"
-                msg = sg.sighash(blob, t, i, ecdsa=ecdsa)
+                ht = ins.get("htype", SIGHASH_ALL)
+                msg = sg.sighash(blob, t, i, hash_type=ht, ecdsa=ecdsa)
                 sig = (sg.ecdsa_sign if ecdsa else sg.schnorr_sign)(ins["key"], msg)
-                sig_script = bytes([0x41]) + sig + bytes([SIGHASH_ALL])
+                sig_script = bytes([0x41]) + sig + bytes([ht])
             if spec["invalid"] and i == 0:
                 sig_script = bytearray(sig_script)
                 sig_script[10] ^= 0x40  # corrupt the first signature

@@ -45,6 +45,8 @@ def engine_validate(engine, blob, n, flags=SKIP_MASS):
           pct_multisig=10, pct_invalid=15), SKIP_MASS),                 # adversarial
     (dict(seed=13, n_txs=30, payload_len=500), SKIP_MASS),              # payloads
     (dict(seed=14, n_txs=60, pct_multi_input=30, pct_ecdsa=10), 0),     # FULL (KIP-9 mass)
+    (dict(seed=16, n_txs=80, pct_multi_input=20, pct_ecdsa=15,
+          pct_alt_hashtype=50), SKIP_MASS),                             # all 6 SigHashTypes
 ])
 def test_validate_block_parity(oracle, engine, kwargs, flags):
     n = kwargs["n_txs"]
