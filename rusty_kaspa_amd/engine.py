@@ -132,10 +132,7 @@ class Engine:
         return list(codes), list(fees), bytes(partial) if want_muhash else None
 
     def sig_cache_stats(self):
-        class _CS(ctypes.Structure):
-            _fields_ = [("insertions", ctypes.c_uint64), ("hits", ctypes.c_uint64),
-                        ("misses", ctypes.c_uint64)]
-        out = _CS()
+        out = KvCacheStats()
         self._check(self.lib.kv_sig_cache_stats(ctypes.c_void_p(self.ctx),
                                                 ctypes.byref(out)))
         return out.insertions, out.hits, out.misses
