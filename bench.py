@@ -110,8 +110,14 @@ def block_mode(args):
     t0 = time.time()
     # each rank validates its own shard of blocks (blocks are independent;
     # the only cross-rank step in the real path is the muhash combine)
-    blob, _ = gen_block(oracle, seed=SEED + rank, n_txs=n_txs, pct_multi_input=20,
-                        pct_ecdsa=10)
+    mix = dict(pct_multi_input=20, pct_ecdsa=10)
+    if args.adversarial:
+        # BASELINE config 5: invalid sigs + large multisig scripts exercising
+        # branch divergence (the 1M-entry UTXO-set leg is the block-utxo mode
+        # with a pre-seeded table; table scaling in tests/test_gpu_utxo.py)
+        mix = dict(pct_multi_input=20, pct_ecdsa=10, pct_multisig=10,
+                   pct_invalid=10)
+    blob, _ = gen_block(oracle, seed=SEED + rank, n_txs=n_txs, **mix)
     log(f"generated {n_txs} mixed txs ({len(blob)/1e6:.1f} MB blob) "
         f"in {time.time()-t0:.1f}s")
     from rusty_kaspa_amd.engine import Engine
@@ -171,7 +177,12 @@ def block_mode(args):
 
     for _ in range(args.warmup):
         codes, _ = one_step()
-    assert all(c == 0 for c in codes), "unexpected invalid txs in bench batch"
+    if args.adversarial:
+        bad = sum(1 for c in codes if c != 0)
+        assert 0 < bad < len(codes), "adversarial mix must reject some txs"
+        log(f"adversarial: {bad}/{len(codes)} txs rejected (expected)")
+    else:
+        assert all(c == 0 for c in codes), "unexpected invalid txs in bench batch"
     if distributed:
         import torch
         import torch.distributed as dist
@@ -221,7 +232,9 @@ def block_mode(args):
         "config": {"workload": ("block-validate-utxo-config3" if args.mode == "block-utxo"
                                 else "block-validate-config3"),
                    "blocks_per_step": args.block_batch, "txs_per_block": 300,
-                   "mix": "70p2pk/20multi-in/10ecdsa", "flags": "FULL", "sig_cache": "off",
+                   "mix": ("60p2pk/20multi-in/10ecdsa/10multisig+10pct-invalid"
+                           if args.adversarial else "70p2pk/20multi-in/10ecdsa"),
+                   "flags": "FULL", "sig_cache": "off",
                    "parallelism": f"dp{world}" if world > 1 else "single"},
         "roofline": None,  # per-kernel rooflines live in the default mode + profiles/
         "cpu_baseline": cpu_baseline,
@@ -246,6 +259,8 @@ def main():
                     default="verify")
     ap.add_argument("--block-batch", type=int, default=16,
                     help="blocks (of 300 txs) per step in --mode block")
+    ap.add_argument("--adversarial", action="store_true",
+                    help="BASELINE config-5 mix: 10%% invalid + multisig")
     args = ap.parse_args()
 
     if args.mode in ("block", "block-utxo"):
