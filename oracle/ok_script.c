@@ -118,6 +118,10 @@ static uint32_t num_ser(int64_t v, uint8_t out[9]) { /* serialize_i64, size=None
 }
 
 static int stk_push_num(stk *s, int64_t v, int metered) {
+  /* SizedEncodeInt<8> serialization (data_stack.rs:127-190): i64::MIN needs a
+   * 9th byte and fails — the reference surfaces it as Serialization error
+   * (mapped like NumberTooBig to UNKNOWN_ERROR in its vector harness) */
+  if (v == INT64_MIN) return KV_SCRIPT_NUMBER_TOO_BIG;
   uint8_t buf[9];
   uint32_t n = num_ser(v, buf);
   return stk_push_copy(s, buf, n, metered);
