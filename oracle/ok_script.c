@@ -866,6 +866,37 @@ static int exec_opcode(eng *E, uint8_t op, const uint8_t *data, uint32_t dlen) {
       free_ents(e, 2);
       return stk_push_copy(&E->d, h, 32, 1);
     }
+    case 0xcd: { /* OpNum2Bin (opcodes/mod.rs:1297-1308) */
+      int32_t size;
+      if ((rc = stk_pop_i32(&E->d, &size))) return rc;
+      if (size < 0) return KV_SCRIPT_INVALID_INDEX; /* i32_to_usize */
+      if (size > 8) return KV_SCRIPT_NOT_MINIMAL_DATA;
+      int64_t v;
+      if ((rc = stk_pop_num(&E->d, &v))) return rc;
+      /* serialize_i64(v, Some(size)) (data_stack.rs:127-162): minimal
+       * magnitude bytes (with saturation byte), error if longer than size,
+       * zero-pad to size, then OR the sign bit into the LAST byte */
+      uint8_t buf[9];
+      uint32_t n = 0;
+      uint64_t mag = v < 0 ? (uint64_t)(-(v + 1)) + 1 : (uint64_t)v;
+      int sat = 0;
+      while (mag) {
+        buf[n] = (uint8_t)(mag & 0xff);
+        sat = (buf[n] & 0x80) != 0;
+        mag >>= 8;
+        n++;
+      }
+      if (sat) buf[n++] = 0;
+      if (n > (uint32_t)size) return KV_SCRIPT_NUMBER_TOO_BIG; /* Serialization */
+      while (n < (uint32_t)size) buf[n++] = 0;
+      if (v < 0) buf[n - 1] |= 0x80;
+      return stk_push_copy(&E->d, buf, n, 1);
+    }
+    case 0xce: { /* OpBin2Num: pop as i64 (<=8B), re-push minimally encoded */
+      int64_t v;
+      if ((rc = stk_pop_num(&E->d, &v))) return rc;
+      return stk_push_num(&E->d, v, 1);
+    }
     default:
       if (op <= 0x4e) { /* data pushes (literal, unmetered) */
         return stk_push_copy(&E->d, data, dlen, 0);
