@@ -168,6 +168,59 @@ static int stk_pop_raw(stk *s, int count, sent out[]) {
   return 0;
 }
 
+/* ---- introspection helpers (opcodes/mod.rs:196-207 substring/i32_to_usize,
+ * tx.rs ScriptPublicKey::to_bytes = version u16 LE + script) ---- */
+static int intro_input(eng *E, const ok_input **out) {
+  int32_t idx;
+  int rc = stk_pop_i32(&E->d, &idx);
+  if (rc) return rc;
+  if (idx < 0) return KV_SCRIPT_INVALID_INDEX;
+  if ((uint32_t)idx >= E->tx->n_inputs) return KV_SCRIPT_INVALID_INPUT_INDEX;
+  *out = &E->tx->inputs[idx];
+  return 0;
+}
+
+static int intro_output(eng *E, const ok_output **out) {
+  int32_t idx;
+  int rc = stk_pop_i32(&E->d, &idx);
+  if (rc) return rc;
+  if (idx < 0) return KV_SCRIPT_INVALID_INDEX;
+  if ((uint32_t)idx >= E->tx->n_outputs) return KV_SCRIPT_INVALID_INPUT_INDEX;
+  *out = &E->tx->outputs[idx];
+  return 0;
+}
+
+static int intro_substr(eng *E, const uint8_t *data, uint32_t len,
+                        int32_t start, int32_t end) {
+  if (start < 0 || end < 0) return KV_SCRIPT_INVALID_INDEX;
+  if (end < start) return KV_SCRIPT_INVALID_RANGE;
+  if ((uint32_t)(end - start) > 1000000u) return KV_SCRIPT_ELEMENT_TOO_BIG;
+  if ((uint32_t)end > len) return KV_SCRIPT_INVALID_SOURCE; /* OutOfBounds */
+  return stk_push_copy(&E->d, data + start, (uint32_t)(end - start), 1);
+}
+
+static int intro_push_spk(eng *E, uint16_t version, const uint8_t *script,
+                          uint32_t len) {
+  uint8_t *buf = (uint8_t *)malloc(2 + (size_t)len);
+  buf[0] = (uint8_t)(version & 0xff);
+  buf[1] = (uint8_t)(version >> 8);
+  memcpy(buf + 2, script, len);
+  int rc = stk_push_copy(&E->d, buf, 2 + len, 1);
+  free(buf);
+  return rc;
+}
+
+static int intro_spk_substr(eng *E, uint16_t version, const uint8_t *script,
+                            uint32_t len, int32_t start, int32_t end) {
+  uint8_t *buf = (uint8_t *)malloc(2 + (size_t)len);
+  buf[0] = (uint8_t)(version & 0xff);
+  buf[1] = (uint8_t)(version >> 8);
+  memcpy(buf + 2, script, len);
+  int rc = intro_substr(E, buf, 2 + len, start, end);
+  free(buf);
+  return rc;
+}
+
 static void free_ents(sent *e, int n) {
   for (int i = 0; i < n; i++) free(e[i].d);
 }
@@ -896,6 +949,217 @@ static int exec_opcode(eng *E, uint8_t op, const uint8_t *data, uint32_t dlen) {
       int64_t v;
       if ((rc = stk_pop_num(&E->d, &v))) return rc;
       return stk_push_num(&E->d, v, 1);
+    }
+    /* ---- transaction introspection (KIP-10 family, opcodes/mod.rs:969-1445).
+     * The oracle always executes in TxInput source context. Index pops follow
+     * i32_to_usize (negative -> InvalidIndex); substring() maps end<start ->
+     * InvalidRange, out-of-bounds -> OutOfBoundsSubstring (both surface as
+     * UNKNOWN_ERROR in the reference's vector harness). */
+    case 0xb2: /* OpTxVersion */
+      return stk_push_num(&E->d, (int64_t)E->tx->version, 1);
+    case 0xb3: /* OpTxInputCount */
+      return stk_push_num(&E->d, (int64_t)E->tx->n_inputs, 1);
+    case 0xb4: /* OpTxOutputCount */
+      return stk_push_num(&E->d, (int64_t)E->tx->n_outputs, 1);
+    case 0xb5: /* OpTxLockTime */
+      return stk_push_num(&E->d, (int64_t)E->tx->lock_time, 1);
+    case 0xb6: /* OpTxSubnetId */
+      return stk_push_copy(&E->d, E->tx->subnetwork_id, 20, 1);
+    case 0xb7: /* OpTxGas */
+      return stk_push_num(&E->d, (int64_t)E->tx->gas, 1);
+    case 0xb8: { /* OpTxPayloadSubstr */
+      int32_t se[2];
+      if ((rc = stk_pop_i32(&E->d, &se[1]))) return rc; /* end on top */
+      if ((rc = stk_pop_i32(&E->d, &se[0]))) return rc;
+      return intro_substr(E, E->tx->payload, E->tx->payload_len, se[0], se[1]);
+    }
+    case 0xb9: /* OpTxInputIndex */
+      return stk_push_num(&E->d, (int64_t)E->idx, 1);
+    case 0xba: { /* OpOutpointTxId */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return stk_push_copy(&E->d, in2->prev_tx_id, 32, 1);
+    }
+    case 0xbb: { /* OpOutpointIndex */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return stk_push_num(&E->d, (int64_t)in2->prev_index, 1);
+    }
+    case 0xbc: { /* OpTxInputScriptSigSubstr */
+      int32_t se[2];
+      if ((rc = stk_pop_i32(&E->d, &se[1]))) return rc;
+      if ((rc = stk_pop_i32(&E->d, &se[0]))) return rc;
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return intro_substr(E, in2->sig_script, in2->sig_script_len, se[0], se[1]);
+    }
+    case 0xbd: { /* OpTxInputSeq: raw 8B LE (bitflag field) */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      uint8_t b[8];
+      for (int i = 0; i < 8; i++) b[i] = (uint8_t)(in2->sequence >> (8 * i));
+      return stk_push_copy(&E->d, b, 8, 1);
+    }
+    case 0xbe: { /* OpTxInputAmount */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      if (in2->utxo_amount > (uint64_t)INT64_MAX) return KV_SCRIPT_NUMBER_TOO_BIG;
+      return stk_push_num(&E->d, (int64_t)in2->utxo_amount, 1);
+    }
+    case 0xbf: { /* OpTxInputSpk: version u16 LE + script */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return intro_push_spk(E, in2->utxo_spk_version, in2->utxo_spk,
+                            in2->utxo_spk_len);
+    }
+    case 0xc0: { /* OpTxInputDaaScore */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return stk_push_num(&E->d, (int64_t)in2->utxo_daa_score, 1);
+    }
+    case 0xc1: { /* OpTxInputIsCoinbase */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return stk_push_num(&E->d, in2->utxo_is_coinbase ? 1 : 0, 1);
+    }
+    case 0xc2: { /* OpTxOutputAmount */
+      const ok_output *o2;
+      if ((rc = intro_output(E, &o2))) return rc;
+      if (o2->value > (uint64_t)INT64_MAX) return KV_SCRIPT_NUMBER_TOO_BIG;
+      return stk_push_num(&E->d, (int64_t)o2->value, 1);
+    }
+    case 0xc3: { /* OpTxOutputSpk */
+      const ok_output *o2;
+      if ((rc = intro_output(E, &o2))) return rc;
+      return intro_push_spk(E, o2->spk_version, o2->spk, o2->spk_len);
+    }
+    case 0xc4: /* OpTxPayloadLen */
+      return stk_push_num(&E->d, (int64_t)E->tx->payload_len, 1);
+    case 0xc5: { /* OpTxInputSpkLen (to_bytes len = 2 + script len) */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return stk_push_num(&E->d, 2 + (int64_t)in2->utxo_spk_len, 1);
+    }
+    case 0xc6: { /* OpTxInputSpkSubstr (over version-prefixed bytes) */
+      int32_t se[2];
+      if ((rc = stk_pop_i32(&E->d, &se[1]))) return rc;
+      if ((rc = stk_pop_i32(&E->d, &se[0]))) return rc;
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return intro_spk_substr(E, in2->utxo_spk_version, in2->utxo_spk,
+                              in2->utxo_spk_len, se[0], se[1]);
+    }
+    case 0xc7: { /* OpTxOutputSpkLen */
+      const ok_output *o2;
+      if ((rc = intro_output(E, &o2))) return rc;
+      return stk_push_num(&E->d, 2 + (int64_t)o2->spk_len, 1);
+    }
+    case 0xc8: { /* OpTxOutputSpkSubstr */
+      int32_t se[2];
+      if ((rc = stk_pop_i32(&E->d, &se[1]))) return rc;
+      if ((rc = stk_pop_i32(&E->d, &se[0]))) return rc;
+      const ok_output *o2;
+      if ((rc = intro_output(E, &o2))) return rc;
+      return intro_spk_substr(E, o2->spk_version, o2->spk, o2->spk_len,
+                              se[0], se[1]);
+    }
+    case 0xc9: { /* OpTxInputScriptSigLen */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      return stk_push_num(&E->d, (int64_t)in2->sig_script_len, 1);
+    }
+    case 0xcb: { /* OpAuthOutputCount (covenants.rs:74) */
+      int32_t ii;
+      if ((rc = stk_pop_i32(&E->d, &ii))) return rc;
+      if (ii < 0) return KV_SCRIPT_INVALID_INDEX;
+      if ((uint32_t)ii >= E->tx->n_inputs) return KV_SCRIPT_INVALID_INPUT_INDEX;
+      int64_t cnt = 0;
+      for (uint32_t o = 0; o < E->tx->n_outputs; o++)
+        if (E->tx->outputs[o].has_covenant &&
+            E->tx->outputs[o].cov_auth_input == (uint16_t)ii)
+          cnt++;
+      return stk_push_num(&E->d, cnt, 1);
+    }
+    case 0xcc: { /* OpAuthOutputIdx (covenants.rs:66) */
+      int32_t k, ii;
+      if ((rc = stk_pop_i32(&E->d, &k))) return rc;
+      if ((rc = stk_pop_i32(&E->d, &ii))) return rc;
+      if (ii < 0 || k < 0) return KV_SCRIPT_INVALID_INDEX;
+      if ((uint32_t)ii >= E->tx->n_inputs) return KV_SCRIPT_INVALID_INPUT_INDEX;
+      int64_t seen = 0;
+      for (uint32_t o = 0; o < E->tx->n_outputs; o++)
+        if (E->tx->outputs[o].has_covenant &&
+            E->tx->outputs[o].cov_auth_input == (uint16_t)ii) {
+          if (seen == k) return stk_push_num(&E->d, (int64_t)o, 1);
+          seen++;
+        }
+      return KV_SCRIPT_INVALID_SOURCE; /* CovenantsError -> UNKNOWN_ERROR */
+    }
+    case 0xcf: { /* OpInputCovenantId (ZERO_HASH when absent) */
+      const ok_input *in2;
+      if ((rc = intro_input(E, &in2))) return rc;
+      uint8_t zero[32] = {0};
+      return stk_push_copy(
+          &E->d, in2->utxo_covenant_id ? in2->utxo_covenant_id : zero, 32, 1);
+    }
+    case 0xd0: case 0xd1: case 0xd2: case 0xd3: { /* OpCov*{Count,Idx} */
+      int32_t k = 0;
+      if (op == 0xd1 || op == 0xd3) {
+        /* stack: [k, covenant_id(top)] — covenant id popped first below */
+      }
+      sent ide;
+      if ((rc = stk_pop_raw(&E->d, 1, &ide))) return rc;
+      if (ide.len != 32) {
+        free(ide.d);
+        return KV_SCRIPT_INVALID_STATE;
+      }
+      if (op == 0xd1 || op == 0xd3) {
+        if ((rc = stk_pop_i32(&E->d, &k))) {
+          free(ide.d);
+          return rc;
+        }
+        if (k < 0) {
+          free(ide.d);
+          return KV_SCRIPT_INVALID_INDEX;
+        }
+      }
+      int64_t seen = 0;
+      int want_inputs = (op == 0xd0 || op == 0xd1);
+      int want_idx = (op == 0xd1 || op == 0xd3);
+      int found = -1;
+      if (want_inputs) {
+        for (uint32_t i = 0; i < E->tx->n_inputs; i++) {
+          const ok_input *in2 = &E->tx->inputs[i];
+          if (in2->utxo_covenant_id && !memcmp(in2->utxo_covenant_id, ide.d, 32)) {
+            if (want_idx && seen == k) found = (int)i;
+            seen++;
+          }
+        }
+      } else {
+        for (uint32_t o = 0; o < E->tx->n_outputs; o++) {
+          const ok_output *o2 = &E->tx->outputs[o];
+          if (o2->has_covenant && !memcmp(o2->cov_id, ide.d, 32)) {
+            if (want_idx && seen == k) found = (int)o;
+            seen++;
+          }
+        }
+      }
+      free(ide.d);
+      if (!want_idx) return stk_push_num(&E->d, seen, 1);
+      if (found < 0) return KV_SCRIPT_INVALID_SOURCE; /* CovenantsError */
+      return stk_push_num(&E->d, (int64_t)found, 1);
+    }
+    case 0xd5: { /* OpOutputCovenantId */
+      const ok_output *o2;
+      if ((rc = intro_output(E, &o2))) return rc;
+      uint8_t zero[32] = {0};
+      return stk_push_copy(&E->d, o2->has_covenant ? o2->cov_id : zero, 32, 1);
+    }
+    case 0xd6: { /* OpOutputAuthorizingInput (-1 when no covenant) */
+      const ok_output *o2;
+      if ((rc = intro_output(E, &o2))) return rc;
+      return stk_push_num(
+          &E->d, o2->has_covenant ? (int64_t)o2->cov_auth_input : -1, 1);
     }
     default:
       if (op <= 0x4e) { /* data pushes (literal, unmetered) */
