@@ -1171,8 +1171,9 @@ static int exec_opcode(eng *E, uint8_t op, const uint8_t *data, uint32_t dlen) {
           op == 0x99)
         return KV_SCRIPT_OPCODE_DISABLED; /* unreachable: checked pre-exec */
       if (op == 0x65 || op == 0x66) return KV_SCRIPT_OPCODE_RESERVED;
-      if (op == 0xa6 || (op >= 0xb2 && op <= 0xc9) || (op >= 0xcb && op <= 0xd6))
-        return KV_SCRIPT_UNSUPPORTED_OPCODE; /* introspection/zk/covenant: round-1 out of scope */
+      if (op == 0xa6 /* OpZkPrecompile */ || op == 0xd4 /* OpChainblockSeqCommit
+             (needs the KIP-21 reachability accessor — out of scope) */)
+        return KV_SCRIPT_UNSUPPORTED_OPCODE;
       return KV_SCRIPT_INVALID_OPCODE; /* 0xca, 0xdb..0xff */
   }
 }
