@@ -168,6 +168,27 @@ static int stk_pop_raw(stk *s, int count, sent out[]) {
   return 0;
 }
 
+
+/* ---- KIP-21 seq-commitment accessor (OpChainblockSeqCommit, mod.rs:1389).
+ * Production queries DAG reachability; the oracle exposes the same shape the
+ * reference's vector harness mocks (lib.rs:2482-2512): one known chain block
+ * mapping to one commitment. Unset (all-zero) => opcode disabled =>
+ * InvalidOpcode, matching a None accessor. */
+static uint8_t g_seqc_block[32];
+static uint8_t g_seqc_commit[32];
+static int g_seqc_set = 0;
+
+void ok_script_set_seq_commit_mock(const uint8_t block32[32],
+                                   const uint8_t commit32[32]) {
+  if (!block32) {
+    g_seqc_set = 0;
+    return;
+  }
+  memcpy(g_seqc_block, block32, 32);
+  memcpy(g_seqc_commit, commit32, 32);
+  g_seqc_set = 1;
+}
+
 /* ---- introspection helpers (opcodes/mod.rs:196-207 substring/i32_to_usize,
  * tx.rs ScriptPublicKey::to_bytes = version u16 LE + script) ---- */
 static int intro_input(eng *E, const ok_input **out) {
@@ -1148,6 +1169,21 @@ static int exec_opcode(eng *E, uint8_t op, const uint8_t *data, uint32_t dlen) {
       if (!want_idx) return stk_push_num(&E->d, seen, 1);
       if (found < 0) return KV_SCRIPT_INVALID_SOURCE; /* CovenantsError */
       return stk_push_num(&E->d, (int64_t)found, 1);
+    }
+    case 0xd4: { /* OpChainblockSeqCommit (KIP-21) */
+      if (!g_seqc_set) return KV_SCRIPT_INVALID_OPCODE; /* accessor None */
+      sent blk;
+      if ((rc = stk_pop_raw(&E->d, 1, &blk))) return rc;
+      if (blk.len != 32) {
+        free(blk.d);
+        return KV_SCRIPT_INVALID_STATE; /* Hash::try_from */
+      }
+      int known = memcmp(blk.d, g_seqc_block, 32) == 0;
+      free(blk.d);
+      /* mock: known block => ancestor with a commitment; anything else =>
+       * is_chain_ancestor None => BlockAlreadyPruned (UNKNOWN_ERROR class) */
+      if (!known) return KV_SCRIPT_INVALID_SOURCE;
+      return stk_push_copy(&E->d, g_seqc_commit, 32, 1);
     }
     case 0xd5: { /* OpOutputCovenantId */
       const ok_output *o2;

@@ -169,3 +169,9 @@ int ok_body_check(const uint8_t *blob, size_t blob_len);
 int ok_validate_mempool(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_score,
                         double feerate_threshold, int threads,
                         int32_t *tx_codes_out, uint64_t *fees_out);
+
+/* test-harness hook: configure the KIP-21 seq-commitment accessor mock
+ * (OpChainblockSeqCommit). NULL block disables (opcode -> InvalidOpcode,
+ * matching a None accessor). */
+void ok_script_set_seq_commit_mock(const uint8_t block32[32],
+                                   const uint8_t commit32[32]);
