@@ -14,8 +14,14 @@ import sys
 sys.path.insert(0, '/root/repo')
 import rusty_kaspa_amd.blob as B
 
+FIXTURES = {
+    'goref_txs.json.gz': 'goref-1060-tx-265-blocks',
+    'goref_pruning_txs.json.gz': 'goref_custom_pruning_depth',
+}
+import os
+NAME = os.environ.get('GOREF_OUT', 'goref_txs.json.gz')
 FIXTURE = ('/root/reference/testing/integration/testdata/dags_for_json_tests/'
-           'goref-1060-tx-265-blocks/blocks.json.gz')
+           f'{FIXTURES[NAME]}/blocks.json.gz')
 O = ctypes.CDLL('/root/repo/oracle/liboracle.so')
 
 with gzip.open(FIXTURE) as f:
@@ -116,9 +122,9 @@ for blob in batches:
         print("codes:", [c for c in codes][:10])
 print(f"oracle: {total} txs, {bad} rejected")
 if bad == 0:
-    data = {"note": "goref-1060-tx-265-blocks real Go-implementation txs with "
+    data = {"note": f"{FIXTURES[NAME]} real Go-implementation txs with "
                     "rebuilt UtxoEntries; every signature must verify",
             "batches": [b.hex() for b in batches]}
-    with gzip.open('/root/repo/tests/golden/goref_txs.json.gz', 'wt') as f:
+    with gzip.open(f'/root/repo/tests/golden/{NAME}', 'wt') as f:
         json.dump(data, f)
-    print("golden written")
+    print("golden written:", NAME)

@@ -14,12 +14,14 @@ import pytest
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", "oracle"))
 
-GOLDEN = os.path.join(os.path.dirname(__file__), "golden", "goref_txs.json.gz")
+GOLD_DIR = os.path.join(os.path.dirname(__file__), "golden")
+FIXTURES = ["goref_txs.json.gz", "goref_pruning_txs.json.gz"]
+EXPECT_TOTAL = {"goref_txs.json.gz": 223, "goref_pruning_txs.json.gz": 4789}
 SKIP_MASS = 2
 
 
-def load_batches():
-    with gzip.open(GOLDEN, "rt") as f:
+def load_batches(name="goref_txs.json.gz"):
+    with gzip.open(os.path.join(GOLD_DIR, name), "rt") as f:
         return [bytes.fromhex(h) for h in json.load(f)["batches"]]
 
 
@@ -34,14 +36,15 @@ def oracle_validate(oracle, blob, n):
     return list(codes), list(fees), bytes(mh)
 
 
-def test_oracle_accepts_all_real_sigs(oracle):
+@pytest.mark.parametrize("name", FIXTURES)
+def test_oracle_accepts_all_real_sigs(oracle, name):
     total = 0
-    for blob in load_batches():
+    for blob in load_batches(name):
         n, = struct.unpack_from("<I", blob, 0)
         codes, fees, _ = oracle_validate(oracle, blob, n)
         assert all(c == 0 for c in codes), codes[:8]
         total += n
-    assert total == 223
+    assert total == EXPECT_TOTAL[name]
 
 
 def test_oracle_rejects_tampered_sig(oracle):
@@ -62,7 +65,8 @@ def test_engine_matches_oracle_on_real_txs(oracle):
     from rusty_kaspa_amd.engine import Engine
     eng = Engine()
     try:
-        for blob in load_batches():
+        for name in FIXTURES:
+          for blob in load_batches(name):
             n, = struct.unpack_from("<I", blob, 0)
             oc, of, omh = oracle_validate(oracle, blob, n)
             ec, ef, ep = eng.validate_block(blob, n, 10**9, 10**9, SKIP_MASS)
