@@ -228,3 +228,12 @@ def test_reference_script_vectors(oracle):
           f"opcode families), {len(failures)} mismatched")
     assert not failures, failures[:10]
     assert ran > 500  # the bulk of the suite must actually run
+
+
+def test_scriptnum_serialize_vectors():
+    """data_stack.rs test_serialize vectors pin ser_i64 (the assembler's
+    number encoding, identical to the engine-side minimal encoding)."""
+    g = json.load(open(os.path.join(GOLD, "scriptnum.json")))
+    assert len(g["cases"]) > 20
+    for c in g["cases"]:
+        assert ser_i64(int(c["num"])).hex() == c["hex"], c
