@@ -27,6 +27,7 @@
 #include "kv_sighash_kernels.hip"
 #include "kv_utxo_kernels.hip"
 #include "kv_validate_host.inc"
+#include "kv_script_host.inc"
 
 static thread_local std::string g_last_error;
 
@@ -197,81 +198,6 @@ extern "C" int kv_verify_ecdsa_batch_status(kv_ctx *ctx, const uint8_t *tuples, 
 
 /* ---------------- MuHash host finalization ---------------- */
 
-/* host keyed blake2b-256 (RFC 7693) — product host code */
-namespace {
-static const uint64_t H_IV[8] = {0x6a09e667f3bcc908ULL, 0xbb67ae8584caa73bULL,
-                                 0x3c6ef372fe94f82bULL, 0xa54ff53a5f1d36f1ULL,
-                                 0x510e527fade682d1ULL, 0x9b05688c2b3e6c1fULL,
-                                 0x1f83d9abfb41bd6bULL, 0x5be0cd19137e2179ULL};
-static const uint8_t H_SIG[12][16] = {
-    {0, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13, 14, 15},
-    {14, 10, 4, 8, 9, 15, 13, 6, 1, 12, 0, 2, 11, 7, 5, 3},
-    {11, 8, 12, 0, 5, 2, 15, 13, 10, 14, 3, 6, 7, 1, 9, 4},
-    {7, 9, 3, 1, 13, 12, 11, 14, 2, 6, 5, 10, 4, 0, 15, 8},
-    {9, 0, 5, 7, 2, 4, 10, 15, 14, 1, 11, 12, 6, 8, 3, 13},
-    {2, 12, 6, 10, 0, 11, 8, 3, 4, 13, 7, 5, 15, 14, 1, 9},
-    {12, 5, 1, 15, 14, 13, 4, 10, 0, 7, 6, 3, 9, 2, 8, 11},
-    {13, 11, 7, 14, 12, 1, 3, 9, 5, 0, 15, 4, 8, 6, 2, 10},
-    {6, 15, 14, 9, 11, 3, 0, 8, 12, 2, 13, 7, 1, 4, 10, 5},
-    {10, 2, 8, 4, 7, 6, 1, 5, 15, 11, 9, 14, 3, 12, 13, 0},
-    {0, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13, 14, 15},
-    {14, 10, 4, 8, 9, 15, 13, 6, 1, 12, 0, 2, 11, 7, 5, 3}};
-
-static inline uint64_t rot64(uint64_t x, int n) { return (x >> n) | (x << (64 - n)); }
-
-static void h_b2b_compress(uint64_t h[8], const uint8_t blk[128], uint64_t t, int last) {
-  uint64_t v[16], m[16];
-  for (int i = 0; i < 16; i++) memcpy(&m[i], blk + 8 * i, 8);
-  for (int i = 0; i < 8; i++) v[i] = h[i];
-  for (int i = 0; i < 8; i++) v[i + 8] = H_IV[i];
-  v[12] ^= t;
-  if (last) v[14] = ~v[14];
-#define HG(a, b, c, d, x, y)                                                   \
-  v[a] += v[b] + (x); v[d] = rot64(v[d] ^ v[a], 32); v[c] += v[d];             \
-  v[b] = rot64(v[b] ^ v[c], 24); v[a] += v[b] + (y);                           \
-  v[d] = rot64(v[d] ^ v[a], 16); v[c] += v[d]; v[b] = rot64(v[b] ^ v[c], 63);
-  for (int r = 0; r < 12; r++) {
-    const uint8_t *s = H_SIG[r];
-    HG(0, 4, 8, 12, m[s[0]], m[s[1]]); HG(1, 5, 9, 13, m[s[2]], m[s[3]]);
-    HG(2, 6, 10, 14, m[s[4]], m[s[5]]); HG(3, 7, 11, 15, m[s[6]], m[s[7]]);
-    HG(0, 5, 10, 15, m[s[8]], m[s[9]]); HG(1, 6, 11, 12, m[s[10]], m[s[11]]);
-    HG(2, 7, 8, 13, m[s[12]], m[s[13]]); HG(3, 4, 9, 14, m[s[14]], m[s[15]]);
-  }
-#undef HG
-  for (int i = 0; i < 8; i++) h[i] ^= v[i] ^ v[i + 8];
-}
-
-static void h_blake2b_keyed(const uint8_t *key, size_t keylen, const uint8_t *data,
-                            size_t len, uint8_t out[32]) {
-  uint64_t h[8];
-  for (int i = 0; i < 8; i++) h[i] = H_IV[i];
-  h[0] ^= 32ULL | ((uint64_t)keylen << 8) | (1ULL << 16) | (1ULL << 24);
-  uint8_t blk[128];
-  uint64_t t = 0;
-  if (keylen) {
-    memset(blk, 0, 128);
-    memcpy(blk, key, keylen);
-    if (len == 0) {
-      h_b2b_compress(h, blk, 128, 1);
-      goto fin;
-    }
-    t = 128;
-    h_b2b_compress(h, blk, 128, 0);
-  }
-  while (len > 128) {
-    t += 128;
-    h_b2b_compress(h, data, t, 0);
-    data += 128;
-    len -= 128;
-  }
-  memset(blk, 0, 128);
-  memcpy(blk, data, len);
-  t += len;
-  h_b2b_compress(h, blk, t, 1);
-fin:
-  for (int i = 0; i < 32; i++) out[i] = (uint8_t)(h[i / 8] >> (8 * (i % 8)));
-}
-
 /* ---- 3072-bit helpers for the modular inverse (binary extended gcd) ---- */
 
 using kv::u3072;
@@ -372,7 +298,6 @@ static void u3072_inverse(const u3072 &in, u3072 &out) {
   memcpy(out.l, big_cmp(u, one) == 0 ? x1 : x2, sizeof(out.l));
 }
 
-} // namespace
 
 extern "C" int kv_muhash_combine(kv_ctx *ctx, uint8_t *acc_partial768,
                                  const uint8_t *other_partial768) {
@@ -401,7 +326,7 @@ extern "C" int kv_muhash_finalize(kv_ctx *ctx, const uint8_t *partial768,
   uint8_t ser[384];
   memcpy(ser, r.l, 384); /* limbs are LE on all supported hosts */
   static const uint8_t KEY[] = "MuHashFinalize";
-  h_blake2b_keyed(KEY, sizeof(KEY) - 1, ser, 384, hash32_out);
+  kvhost::h_blake2b_keyed(KEY, sizeof(KEY) - 1, ser, 384, hash32_out);
   return 0;
 }
 
@@ -502,6 +427,22 @@ struct ValidateBufs {
 static ValidateBufs g_vb; /* guarded by ctx->mu (single validate at a time) */
 
 /* classify one input; appends jobs. Returns plan. */
+/* non-template scripts: run the host general interpreter (kv_script_host.inc).
+ * Scripts whose execution reaches a signature-check opcode DEFER and keep the
+ * round-1 UNSUPPORTED routing (GPU owns EC verification); everything else —
+ * hash puzzles, timelocks, introspection/covenant logic, anyone-can-spend —
+ * resolves right here. */
+static int32_t unsupported_pre_code(const HTx &tx, const HInput &in,
+                                    uint32_t input_index, uint64_t sigop_units) {
+  int rc = kvh_run_input_script(tx, in, input_index, sigop_units);
+  int base = in.sig_script_len == 0 ? KV_ERR_SIGNATURE_EMPTY_BASE
+                                    : KV_ERR_SIGNATURE_INVALID_BASE;
+  if (rc == kvhost::KVH_SCRIPT_DEFER)
+    return base + KV_SCRIPT_UNSUPPORTED_OPCODE;
+  if (rc == 0) return 0;
+  return base + rc;
+}
+
 static InputPlan classify_input(const HTx &tx, const HInput &in,
                                 uint64_t sigop_units,
                                 std::vector<kv::kv_job> &sjobs,
@@ -539,7 +480,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
       return pl;
     }
     if (in.sig_script_len != 1 + siglen || in.sig_script[0] != siglen) {
-      pl.pre_code = KV_ERR_SIGNATURE_INVALID_BASE + KV_SCRIPT_UNSUPPORTED_OPCODE;
+      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
       return pl;
     }
     uint8_t type = in.sig_script[siglen];
@@ -582,7 +523,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     if (!parse_pushes(nullptr, in.sig_script, in.sig_script_len, in.sig_script_off,
                       pushes) ||
         pushes.empty()) {
-      pl.pre_code = KV_ERR_SIGNATURE_INVALID_BASE + KV_SCRIPT_UNSUPPORTED_OPCODE;
+      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
       return pl;
     }
     /* last push = redeem candidate (need the actual bytes: offset into blob is
@@ -607,7 +548,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     pl.spent_units += 1; /* OpEqual pushed [1] */
     /* parse canonical multisig redeem: OP_m (0x20 pk)×n OP_n 0xae */
     if (rlen < 3 || redeem[0] < 0x51 || redeem[0] > 0x60) {
-      pl.pre_code = KV_ERR_SIGNATURE_INVALID_BASE + KV_SCRIPT_UNSUPPORTED_OPCODE;
+      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
       return pl;
     }
     int m = redeem[0] - 0x50;
@@ -620,7 +561,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     int n = (int)key_offs.size();
     if (rp + 2 != rlen || n < 1 || n > 20 || redeem[rp] != (uint8_t)(0x50 + n) ||
         redeem[rp + 1] != 0xae || m > n || m < 1) {
-      pl.pre_code = KV_ERR_SIGNATURE_INVALID_BASE + KV_SCRIPT_UNSUPPORTED_OPCODE;
+      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
       return pl;
     }
     int nsigs = (int)pushes.size() - 1;
@@ -666,9 +607,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     return pl;
   }
 
-  pl.pre_code = (in.sig_script_len == 0 ? KV_ERR_SIGNATURE_EMPTY_BASE
-                                        : KV_ERR_SIGNATURE_INVALID_BASE) +
-                KV_SCRIPT_UNSUPPORTED_OPCODE;
+  pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
   return pl;
 }
 
