@@ -112,3 +112,49 @@ def test_malformed_blob_fails_cleanly(oracle, engine):
             engine.validate_block(blob[:cut], 8, 10**9, 10**9, SKIP_MASS)
     codes, _, _ = engine.validate_block(blob, 8, 10**9, 10**9, SKIP_MASS)
     assert all(c == 0 for c in codes)
+
+
+def test_non_template_script_fallback(oracle, engine):
+    """A signature-free non-template script (sha256 hash puzzle) resolves
+    through the host general interpreter inside classify (kv_script_host.inc)
+    — engine codes/fees stay oracle-bit-exact for solve and fail cases."""
+    import hashlib
+    import struct
+    n = 12
+    blob, _ = gen_block(oracle, seed=17, n_txs=n)  # all 1-input P2PK txs
+
+    def with_puzzle(blob, preimage):
+        # rewrite tx 0's only input: spk = OpSHA256 <h32> OpEqual,
+        # sig_script = push(preimage). 1-input tx → no signatures anywhere.
+        b = bytearray(blob)
+        n_txs, = struct.unpack_from("<I", b, 0)
+        off, = struct.unpack_from("<I", b, 4)
+        payload_len, = struct.unpack_from("<I", b, off + 36)
+        p = off + 88 + payload_len  # input record
+        sig_len, = struct.unpack_from("<I", b, p + 48)
+        h = hashlib.sha256(b"secret").digest()
+        new_sig = bytes([len(preimage)]) + preimage
+        new_spk = bytes([0xA8, 0x20]) + h + bytes([0x87])
+        entry_off = p + 52 + sig_len
+        spk_len, = struct.unpack_from("<I", b, entry_off + 20)
+        rest = bytes(b[entry_off + 24 + spk_len:])
+        head = bytes(b[:p + 48])
+        mid = bytes(b[entry_off:entry_off + 20])
+        out = bytearray()
+        out += head
+        out += struct.pack("<I", len(new_sig)) + new_sig
+        out += mid + struct.pack("<I", len(new_spk)) + new_spk
+        out += rest
+        # fix the offset table for the shifted following txs
+        delta = len(out) - len(b)
+        for t in range(1, n_txs):
+            o, = struct.unpack_from("<I", bytes(out), 4 + 4 * t)
+            struct.pack_into("<I", out, 4 + 4 * t, o + delta)
+        return bytes(out)
+
+    for preimage, expect0 in [(b"secret", 0), (b"wrong!", 101)]:
+        pb = with_puzzle(blob, preimage)
+        oc, of, omh = oracle_validate(oracle, pb, n)
+        assert oc[0] == expect0, (oc[0], expect0)
+        ec, ef, emh = engine_validate(engine, pb, n)
+        assert ec == oc and ef == of and emh == omh
