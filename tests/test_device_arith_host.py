@@ -46,7 +46,7 @@ def val(a):
 def test_fe26_mul_sqr_vs_bigint(shim):
     rng = random.Random(11)
     arr = lambda l: (ctypes.c_uint32 * 10)(*l)
-    for _ in range(4000):
+    for _ in range(8000):
         m = rng.choice([1, 2, 4, 8])
         a, b = rnd_limbs(rng, m), rnd_limbs(rng, m)
         r = (ctypes.c_uint32 * 10)()

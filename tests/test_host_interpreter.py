@@ -86,7 +86,7 @@ def test_random_script_fuzz(oracle, shim):
     rng = random.Random(99)
     agreed = deferred = 0
     mismatches = []
-    for trial in range(4000):
+    for trial in range(12000):
         slen = rng.randrange(0, 40)
         spk = bytes(rng.randrange(256) for _ in range(slen))
         sig_len = rng.randrange(0, 30)
