@@ -59,12 +59,14 @@ def load_oracle():
     return ctypes.CDLL(so)
 
 
-def gen_tuples(oracle, n, seed, invalid_permille):
+def gen_tuples(oracle, n, seed, invalid_permille, world=1):
     buf = ctypes.create_string_buffer(n * 128)
     t0 = time.time()
+    # N ranks generate concurrently on one host: share the cores
+    threads = max(8, (os.cpu_count() or 8) // max(1, world))
     oracle.ok_gen_schnorr_tuples(ctypes.c_uint64(seed), ctypes.c_size_t(n),
                                  ctypes.c_uint32(invalid_permille), buf,
-                                 os.cpu_count() or 8)
+                                 threads)
     log(f"generated {n} tuples in {time.time()-t0:.1f}s (seed {seed})")
     return buf
 
@@ -280,7 +282,7 @@ def main():
     oracle = load_oracle()
     n = args.tuples
     # each rank gets its own deterministic batch (round-robin shard of the stream)
-    tuples = gen_tuples(oracle, n, SEED + rank, args.invalid_permille)
+    tuples = gen_tuples(oracle, n, SEED + rank, args.invalid_permille, world)
 
     from rusty_kaspa_amd.engine import Engine
     eng = Engine(device=local_rank, sig_cache_size=0)
