@@ -100,3 +100,18 @@ extern "C" void host_fe26_neg_norm(const uint32_t a[10], uint32_t m, uint32_t r[
   kv::fe26_normalize(fr);
   for (int i = 0; i < 10; i++) r[i] = fr.l[i];
 }
+extern "C" void host_fe26_inv(const uint32_t a[10], uint32_t r[10]) {
+  kv::fe26 fa, fr;
+  for (int i = 0; i < 10; i++) fa.l[i] = a[i];
+  kv::fe26_inv(fr, fa);
+  kv::fe26_normalize(fr);
+  for (int i = 0; i < 10; i++) r[i] = fr.l[i];
+}
+extern "C" int host_fe26_sqrt(const uint32_t a[10], uint32_t r[10]) {
+  kv::fe26 fa, fr;
+  for (int i = 0; i < 10; i++) fa.l[i] = a[i];
+  int ok = kv::fe26_sqrt(fr, fa);
+  kv::fe26_normalize(fr);
+  for (int i = 0; i < 10; i++) r[i] = fr.l[i];
+  return ok;
+}
