@@ -103,3 +103,21 @@ def test_schnorr_ecdsa_vs_oracle(shim, oracle):
         got = shim.host_ecdsa_verify(t[64:97], t[97:129], t[:64])
         expd = oracle.ok_ecdsa_verify(t[64:97], t[97:129], t[:64])
         assert got == expd, i
+
+
+def test_fe26_inv_sqrt_vs_pow(shim):
+    """Addition-chain Fermat powers (fe26_inv, fe26_sqrt) against python pow."""
+    rng = random.Random(55)
+    arr = lambda l: (ctypes.c_uint32 * 10)(*l)
+    M26 = (1 << 26) - 1
+    to26 = lambda v: [(v >> (26 * i)) & M26 for i in range(10)]
+    for _ in range(300):
+        v = rng.randrange(1, P)
+        r = (ctypes.c_uint32 * 10)()
+        shim.host_fe26_inv(arr(to26(v)), r)
+        assert val(r) == pow(v, P - 2, P)
+        ok = shim.host_fe26_sqrt(arr(to26(v)), r)
+        is_qr = pow(v, (P - 1) // 2, P) == 1
+        assert bool(ok) == is_qr
+        if ok:
+            assert (val(r) * val(r)) % P == v
