@@ -60,7 +60,14 @@ __device__ __forceinline__ int key_eq(const uint8_t *a, const uint8_t *b) {
   return x == 0 && ia == ib;
 }
 
-/* one outpoint per lane; values are prepacked 64B records */
+/* one outpoint per lane; values are prepacked 64B records.
+ *
+ * Tombstone discipline: an upsert must probe its FULL chain (until an EMPTY
+ * slot or a key match) before claiming anything — a tombstone left early in a
+ * live key's chain by a prior remove must not be claimed until the key is
+ * known absent, or the table would hold two READY slots for one outpoint and a
+ * later remove would resurrect the stale one. We remember the first tombstone
+ * seen and claim it only after reaching the chain's end without a match. */
 extern "C" __global__ void kv_utxo_upsert_kernel(utxo_slot *table, uint64_t cap_mask,
                                                  const uint8_t *__restrict__ outpoints,
                                                  const uint8_t *__restrict__ values,
@@ -70,37 +77,53 @@ extern "C" __global__ void kv_utxo_upsert_kernel(utxo_slot *table, uint64_t cap_
   if (i >= n) return;
   const uint8_t *key = outpoints + i * 36;
   const uint8_t *val = values + i * 64;
-  uint64_t slot = op_hash(key) & cap_mask;
-  for (uint64_t probe = 0; probe <= cap_mask; probe++, slot = (slot + 1) & cap_mask) {
-    utxo_slot *s = &table[slot];
-    uint32_t st = __hip_atomic_load(&s->state, __ATOMIC_ACQUIRE,
-                                    __HIP_MEMORY_SCOPE_AGENT);
-    if (st == KV_SLOT_READY || st == KV_SLOT_CLAIMED) {
+  const uint64_t home = op_hash(key) & cap_mask;
+  for (;;) { /* restart only on a lost claim race (rare) */
+    uint64_t cand = ~0ULL; /* first tombstone seen this scan */
+    uint64_t slot = home;
+    int restart = 0;
+    uint64_t probe = 0;
+    for (; probe <= cap_mask; probe++, slot = (slot + 1) & cap_mask) {
+      utxo_slot *s = &table[slot];
+      uint32_t st = __hip_atomic_load(&s->state, __ATOMIC_ACQUIRE,
+                                      __HIP_MEMORY_SCOPE_AGENT);
       /* wait for a concurrent claimer of this slot to publish its key */
       while (st == KV_SLOT_CLAIMED)
         st = __hip_atomic_load(&s->state, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
-      if (key_eq(s->key, key)) { /* overwrite value (upsert semantics) */
-        for (int k = 0; k < 64; k++) s->value[k] = val[k];
-        return;
+      if (st == KV_SLOT_READY) {
+        if (key_eq(s->key, key)) { /* overwrite value (upsert semantics) */
+          for (int k = 0; k < 64; k++) s->value[k] = val[k];
+          return;
+        }
+        continue;
       }
-      continue;
+      if (st == KV_SLOT_TOMB) {
+        if (cand == ~0ULL) cand = slot;
+        continue; /* keep probing: the key may live further down the chain */
+      }
+      /* EMPTY: end of this key's chain — the key is absent. Claim the
+       * remembered tombstone if any, else this empty slot. */
+      {
+        uint64_t tgt = (cand != ~0ULL) ? cand : slot;
+        uint32_t expected = (cand != ~0ULL) ? KV_SLOT_TOMB : KV_SLOT_EMPTY;
+        utxo_slot *t = &table[tgt];
+        if (__hip_atomic_compare_exchange_strong(&t->state, &expected, KV_SLOT_CLAIMED,
+                                                 __ATOMIC_ACQ_REL, __ATOMIC_ACQUIRE,
+                                                 __HIP_MEMORY_SCOPE_AGENT)) {
+          for (int k = 0; k < 36; k++) t->key[k] = key[k];
+          for (int k = 0; k < 64; k++) t->value[k] = val[k];
+          __hip_atomic_store(&t->state, KV_SLOT_READY, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          return;
+        }
+        restart = 1; /* another lane took the slot: rescan from home */
+        break;
+      }
     }
-    /* empty or tombstone: try to claim */
-    uint32_t expected = st;
-    if (__hip_atomic_compare_exchange_strong(&s->state, &expected, KV_SLOT_CLAIMED,
-                                             __ATOMIC_ACQ_REL, __ATOMIC_ACQUIRE,
-                                             __HIP_MEMORY_SCOPE_AGENT)) {
-      for (int k = 0; k < 36; k++) s->key[k] = key[k];
-      for (int k = 0; k < 64; k++) s->value[k] = val[k];
-      __hip_atomic_store(&s->state, KV_SLOT_READY, __ATOMIC_RELEASE,
-                         __HIP_MEMORY_SCOPE_AGENT);
-      return;
-    }
-    /* lost the race: re-examine this slot (it may now hold our key) */
-    probe--;
-    slot = (slot - 1) & cap_mask;
+    if (restart) continue;
+    if (fail_flag) *fail_flag = 1; /* table full */
+    return;
   }
-  if (fail_flag) *fail_flag = 1; /* table full */
 }
 
 extern "C" __global__ void kv_utxo_remove_kernel(utxo_slot *table, uint64_t cap_mask,

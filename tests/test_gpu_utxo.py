@@ -118,3 +118,63 @@ def test_utxo_1m_scale_and_throughput(engine):
     assert all(found), found.count(0)
     print(f"\n[utxo] {batch} random lookups over 1M-entry table: {ms:.2f} ms "
           f"= {batch / ms * 1000 / 1e6:.1f}M lookups/s")
+
+
+def _op_home(op36, mask):
+    """Python model of kv_utxo_kernels.hip op_hash()."""
+    h = int.from_bytes(op36[:8], "little")
+    idx = int.from_bytes(op36[32:36], "little")
+    return (h ^ (0x9E3779B97F4A7C15 * (idx + 1) & (1 << 64) - 1)) & mask
+
+
+def test_utxo_tombstone_chain_upsert(engine):
+    """Regression (advisor r1): a tombstone left early in a live key's probe
+    chain must not be claimed by an upsert of that key — the upsert must keep
+    probing, find the existing READY slot and overwrite it, or the table holds
+    two slots for one outpoint and a later remove resurrects the stale one."""
+    lib = engine.lib
+    assert lib.kv_utxo_reset(ctypes.c_void_p(engine.ctx), ctypes.c_uint64(16)) == 0
+    mask = 63  # reset(16) allocates the 64-slot minimum
+
+    # brute-force three distinct outpoints sharing one home slot
+    rng = random.Random(99)
+    home, cluster = None, []
+    while len(cluster) < 3:
+        op = outpoint(rng)
+        h = _op_home(op, mask)
+        if home is None:
+            home, cluster = h, [op]
+        elif h == home and op not in cluster:
+            cluster.append(op)
+    k1, k2, k3 = cluster
+
+    def upsert(op, ent):
+        rc = lib.kv_utxo_upsert(ctypes.c_void_p(engine.ctx), op, ent,
+                                ctypes.c_size_t(1))
+        assert rc == 0, lib.kv_last_error().decode()
+
+    def remove(op):
+        assert lib.kv_utxo_remove(ctypes.c_void_p(engine.ctx), op,
+                                  ctypes.c_size_t(1)) == 0
+
+    e_old = pack_entry(111, 1, False, b"\x01" * 34)
+    e_new = pack_entry(222, 2, False, b"\x02" * 34)
+    e_k1 = pack_entry(333, 3, False, b"\x03" * 34)
+
+    upsert(k1, e_k1)   # lands at home
+    upsert(k2, e_old)  # chains to home+1
+    remove(k1)         # tombstone at home, k2's chain passes through it
+    upsert(k2, e_new)  # MUST overwrite home+1, not claim the tombstone
+
+    found, entries, _ = lookup(engine, [k2])
+    assert found == [1]
+    assert entries[:64] == e_new
+
+    remove(k2)         # must delete the ONE slot holding k2
+    found, _, _ = lookup(engine, [k1, k2])
+    assert found == [0, 0], "stale k2 entry resurrected after remove"
+
+    # a fresh key may still reuse the tombstones
+    upsert(k3, e_old)
+    found, entries, _ = lookup(engine, [k3])
+    assert found == [1] and entries[:64] == e_old
