@@ -15,6 +15,7 @@
  */
 #include "kaspa_engine_abi.h"
 #include "ok_tx.h"
+#include <math.h>
 #include <string.h>
 #ifdef _OPENMP
 #include <omp.h>
@@ -381,6 +382,22 @@ static uint64_t mp_estimated_size(const ok_tx *tx) {
   return size;
 }
 
+/* Generic normalized_max over per-dimension limits ⇔ Mass::normalized_max +
+ * MassCofactors::new (mass/mod.rs:258-265,298-308): cofactor_i = Lc/L_i (f64),
+ * normalized = ceil(m_i · cofactor_i), result = max over dimensions. */
+uint64_t ok_normalized_max_limits(uint64_t storage_mass, uint64_t compute_mass,
+                                  uint64_t transient_mass, uint64_t limit_storage,
+                                  uint64_t limit_compute, uint64_t limit_transient) {
+  double cs = (double)limit_compute / (double)limit_storage;
+  double ct = (double)limit_compute / (double)limit_transient;
+  uint64_t sn = (uint64_t)ceil((double)storage_mass * cs);
+  uint64_t tn = (uint64_t)ceil((double)transient_mass * ct);
+  uint64_t m = sn;
+  if (compute_mass > m) m = compute_mass;
+  if (tn > m) m = tn;
+  return m;
+}
+
 static uint64_t mp_normalized_mass(const ok_tx *tx, uint64_t storage_mass) {
   if (ok_tx_is_coinbase(tx)) return storage_mass; /* non-contextual = 0 */
   uint64_t size = mp_estimated_size(tx);
@@ -401,12 +418,9 @@ static uint64_t mp_normalized_mass(const ok_tx *tx, uint64_t storage_mass) {
   }
   uint64_t compute = size * 1 + spk_bytes * 10 + script_mass;
   uint64_t transient = size * 4;
-  /* normalized_max: max(ceil(storage*1.0), compute, ceil(transient*0.5)) */
-  uint64_t tnorm = (transient + 1) / 2;
-  uint64_t m = storage_mass;
-  if (compute > m) m = compute;
-  if (tnorm > m) m = tnorm;
-  return m;
+  /* mainnet limits {storage 500k, compute 500k, transient 1M}, params.rs:626 */
+  return ok_normalized_max_limits(storage_mass, compute, transient, 500000,
+                                  500000, 1000000);
 }
 
 int ok_validate_mempool(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_score,
@@ -438,4 +452,46 @@ int ok_validate_mempool(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_s
     ok_tx_free(&tx);
   }
   return 0;
+}
+
+/* ---- direct test exports for the golden mass vectors (tests/golden/mass.json,
+ * extracted from consensus/core/src/mass/mod.rs:531-953). Test infrastructure:
+ * they expose the same static helpers the validation path uses. ---- */
+
+uint64_t ok_test_plurality(uint32_t spk_len, int has_cov) {
+  return plurality(spk_len, has_cov);
+}
+
+/* Build a minimal non-coinbase tx view carrying only the fields
+ * calc_storage_mass_tx reads. Returns 0 ok (mass in *out), -1 incomputable. */
+int ok_test_storage_mass(uint32_t n_ins, const uint64_t *in_amounts,
+                         const uint32_t *in_spk_lens, const uint8_t *in_has_cov,
+                         uint32_t n_outs, const uint64_t *out_amounts,
+                         const uint32_t *out_spk_lens, const uint8_t *out_has_cov,
+                         uint64_t *mass_out) {
+  static const uint8_t zero_subnet[20] = {0};
+  static const uint8_t cov_stub[32] = {0};
+  ok_input ins[64];
+  ok_output outs[64];
+  if (n_ins > 64 || n_outs > 64) return -2;
+  memset(ins, 0, sizeof(ins[0]) * n_ins);
+  memset(outs, 0, sizeof(outs[0]) * n_outs);
+  for (uint32_t i = 0; i < n_ins; i++) {
+    ins[i].utxo_amount = in_amounts[i];
+    ins[i].utxo_spk_len = in_spk_lens ? in_spk_lens[i] : 0;
+    ins[i].utxo_covenant_id = (in_has_cov && in_has_cov[i]) ? cov_stub : NULL;
+  }
+  for (uint32_t i = 0; i < n_outs; i++) {
+    outs[i].value = out_amounts[i];
+    outs[i].spk_len = out_spk_lens ? out_spk_lens[i] : 0;
+    outs[i].cov_id = (out_has_cov && out_has_cov[i]) ? cov_stub : NULL;
+  }
+  ok_tx tx;
+  memset(&tx, 0, sizeof(tx));
+  tx.n_inputs = n_ins;
+  tx.n_outputs = n_outs;
+  tx.subnetwork_id = zero_subnet;
+  tx.inputs = ins;
+  tx.outputs = outs;
+  return calc_storage_mass_tx(&tx, mass_out);
 }

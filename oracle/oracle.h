@@ -175,3 +175,18 @@ int ok_validate_mempool(const uint8_t *blob, size_t blob_len, uint64_t pov_daa_s
  * matching a None accessor). */
 void ok_script_set_seq_commit_mock(const uint8_t block32[32],
                                    const uint8_t commit32[32]);
+
+/* ---- direct mass-vector exports (tests/golden/mass.json) ----
+ * utxo_plurality (mass/mod.rs:83-105) */
+uint64_t ok_test_plurality(uint32_t spk_len, int has_cov);
+/* calc_storage_mass (mass/mod.rs:439-514) over bare (amount, spk_len, has_cov)
+ * cells; 0 ok (mass in *mass_out), -1 incomputable, -2 too many cells */
+int ok_test_storage_mass(uint32_t n_ins, const uint64_t *in_amounts,
+                         const uint32_t *in_spk_lens, const uint8_t *in_has_cov,
+                         uint32_t n_outs, const uint64_t *out_amounts,
+                         const uint32_t *out_spk_lens, const uint8_t *out_has_cov,
+                         uint64_t *mass_out);
+/* Mass::normalized_max + MassCofactors::new (mass/mod.rs:258-265,298-308) */
+uint64_t ok_normalized_max_limits(uint64_t storage_mass, uint64_t compute_mass,
+                                  uint64_t transient_mass, uint64_t limit_storage,
+                                  uint64_t limit_compute, uint64_t limit_transient);

@@ -1724,3 +1724,44 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
   }
   return 0;
 }
+
+/* ---- direct mass-vector test exports (tests/golden/mass.json, extracted from
+ * consensus/core/src/mass/mod.rs:531-953). These expose the engine's OWN mass
+ * helpers (kv_validate_host.inc) so the CPU suite can pin them against the
+ * reference's vectors without a GPU. ---- */
+
+extern "C" uint64_t kv_test_plurality(uint32_t spk_len, int has_cov) {
+  return kvh_plurality(spk_len, has_cov != 0);
+}
+
+extern "C" uint64_t kv_test_normalized_max(uint64_t storage_mass, uint64_t compute_mass,
+                                           uint64_t transient_mass, uint64_t limit_storage,
+                                           uint64_t limit_compute,
+                                           uint64_t limit_transient) {
+  return kvh_normalized_max_limits(storage_mass, compute_mass, transient_mass,
+                                   limit_storage, limit_compute, limit_transient);
+}
+
+extern "C" int kv_test_storage_mass(uint32_t n_ins, const uint64_t *in_amounts,
+                                    const uint32_t *in_spk_lens, const uint8_t *in_has_cov,
+                                    uint32_t n_outs, const uint64_t *out_amounts,
+                                    const uint32_t *out_spk_lens,
+                                    const uint8_t *out_has_cov, uint64_t *mass_out) {
+  HTx tx;
+  tx.version = 1;
+  for (uint32_t i = 0; i < n_ins; i++) {
+    HInput in{};
+    in.utxo_amount = in_amounts[i];
+    in.utxo_spk_len = in_spk_lens ? in_spk_lens[i] : 0;
+    in.utxo_has_cov = (in_has_cov && in_has_cov[i]) ? 1 : 0;
+    tx.inputs.push_back(in);
+  }
+  for (uint32_t i = 0; i < n_outs; i++) {
+    HOutput o{};
+    o.value = out_amounts[i];
+    o.spk_len = out_spk_lens ? out_spk_lens[i] : 0;
+    o.has_covenant = (out_has_cov && out_has_cov[i]) ? 1 : 0;
+    tx.outputs.push_back(o);
+  }
+  return kvh_storage_mass(tx, /*is_coinbase=*/false, mass_out);
+}
