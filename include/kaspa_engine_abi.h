@@ -128,6 +128,14 @@ extern "C" {
 #define KV_SCRIPT_INVALID_RANGE 33
 #define KV_SCRIPT_UNSUPPORTED_OPCODE 63 /* engine-only: routed to CPU fallback */
 
+/* Structurally-distinct "route this tx back to the reference CPU interpreter"
+ * signal: an engine per-tx code that is NEGATIVE, so no binder can fold it
+ * into the reject path by accident (every reject code is >= 1; 0 is accept).
+ * Emitted when a script's execution reaches an opcode the engine does not
+ * decide (currently only OpZkPrecompile, 0xa6). The legacy 1xx/2xx+63 codes
+ * are no longer produced by the engine. */
+#define KV_TX_DEFER_TO_CPU (-63)
+
 /* ---- engine lifecycle ---- */
 
 typedef struct kv_ctx kv_ctx; /* opaque; thread-safe (Sync), internally synchronized */
@@ -253,6 +261,23 @@ int kv_validate_mempool(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
  * chained tx). Either out-pointer may be NULL. */
 int kv_block_body_check(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
                         uint8_t merkle_root_out[32], int32_t *rule_code_out);
+
+/* ---- KIP-21 sequencing-commitment accessor ----
+ * ⇔ SeqCommitAccessor consulted by OpChainblockSeqCommit
+ * (crypto/txscript/src/opcodes/mod.rs:1389-1405; threaded through
+ * tx_validation_in_utxo_context.rs:44,170). The host (Rust over FFI) supplies
+ * a callback so DAG reachability stays on the node side. Contract:
+ *   return 0  → block is a chain ancestor within depth; write the 32-byte
+ *               commitment to commitment_out32
+ *   return nonzero → not an ancestor / pruned / too deep (the script fails
+ *               with the InvalidSource class, matching the reference's
+ *               BlockNotSelected/BlockAlreadyPruned/BlockIsTooDeep rejects)
+ * A NULL accessor (the default) disables the opcode: executing it fails with
+ * InvalidOpcode, matching a None accessor in the reference.
+ * The callback may be invoked concurrently from engine worker threads. */
+typedef int (*kv_seq_commit_accessor_fn)(void *user, const uint8_t block_hash32[32],
+                                         uint8_t commitment_out32[32]);
+int kv_set_seq_commit_accessor(kv_ctx *ctx, kv_seq_commit_accessor_fn fn, void *user);
 
 /* Sig-cache statistics (crypto/txscript/src/caches.rs:57-82 counters). */
 typedef struct {

@@ -59,6 +59,9 @@ struct kv_ctx {
   kv_params params;
   hipStream_t stream;
   std::mutex mu;
+  /* KIP-21 seq-commitment accessor (kv_set_seq_commit_accessor) */
+  kv_seq_commit_accessor_fn seqc_fn = nullptr;
+  void *seqc_user = nullptr;
   /* grow-on-demand device scratch */
   uint8_t *d_in = nullptr;
   size_t d_in_cap = 0;
@@ -330,6 +333,14 @@ extern "C" int kv_muhash_finalize(kv_ctx *ctx, const uint8_t *partial768,
   return 0;
 }
 
+extern "C" int kv_set_seq_commit_accessor(kv_ctx *ctx, kv_seq_commit_accessor_fn fn,
+                                          void *user) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  ctx->seqc_fn = fn;
+  ctx->seqc_user = user;
+  return 0;
+}
+
 extern "C" int kv_sig_cache_stats(kv_ctx *ctx, kv_cache_stats *out) {
   out->insertions = ctx->cache_insertions;
   out->hits = ctx->cache_hits;
@@ -427,23 +438,37 @@ struct ValidateBufs {
 static ValidateBufs g_vb; /* guarded by ctx->mu (single validate at a time) */
 
 /* classify one input; appends jobs. Returns plan. */
-/* non-template scripts: run the host general interpreter (kv_script_host.inc).
- * Scripts whose execution reaches a signature-check opcode DEFER and keep the
- * round-1 UNSUPPORTED routing (GPU owns EC verification); everything else —
- * hash puzzles, timelocks, introspection/covenant logic, anyone-can-spend —
- * resolves right here. */
-static int32_t unsupported_pre_code(const HTx &tx, const HInput &in,
-                                    uint32_t input_index, uint64_t sigop_units) {
-  int rc = kvh_run_input_script(tx, in, input_index, sigop_units);
+/* non-template scripts: run the host general interpreter (kv_script_host.inc)
+ * in collect mode. Signature-free scripts — hash puzzles, timelocks,
+ * introspection/covenant logic, anyone-can-spend — resolve right here; a
+ * script whose execution reaches a signature site suspends with its verify
+ * requests in plan.pending and resolves over the GPU batch in the
+ * interpreter-rounds phase of validate_block_impl. Only zk-precompile scripts
+ * defer to the reference CPU interpreter (KV_TX_DEFER_TO_CPU). */
+static void interp_classify(kv_ctx *ctx, const HTx &tx, const HInput &in,
+                            uint32_t input_index, uint64_t sigop_units,
+                            InputPlan &pl) {
+  kvhost::KvsRunCtx rctx;
+  rctx.memo = &pl.memo;
+  rctx.pending = &pl.pending;
+  rctx.seqc_fn = ctx->seqc_fn;
+  rctx.seqc_user = ctx->seqc_user;
+  int rc = kvh_run_input_script(tx, in, input_index, sigop_units, &rctx);
+  if (rc == kvhost::KVH_SCRIPT_SUSPEND) {
+    pl.kind = PLAN_INTERP;
+    return;
+  }
   int base = in.sig_script_len == 0 ? KV_ERR_SIGNATURE_EMPTY_BASE
                                     : KV_ERR_SIGNATURE_INVALID_BASE;
   if (rc == kvhost::KVH_SCRIPT_DEFER)
-    return base + KV_SCRIPT_UNSUPPORTED_OPCODE;
-  if (rc == 0) return 0;
-  return base + rc;
+    pl.pre_code = KV_TX_DEFER_TO_CPU;
+  else if (rc != 0)
+    pl.pre_code = base + rc;
+  else
+    pl.pre_code = 0;
 }
 
-static InputPlan classify_input(const HTx &tx, const HInput &in,
+static InputPlan classify_input(kv_ctx *ctx, const HTx &tx, const HInput &in,
                                 uint64_t sigop_units,
                                 std::vector<kv::kv_job> &sjobs,
                                 std::vector<kv::kv_job> &ejobs,
@@ -480,7 +505,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
       return pl;
     }
     if (in.sig_script_len != 1 + siglen || in.sig_script[0] != siglen) {
-      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
+      interp_classify(ctx, tx, in, input_index, sigop_units, pl);
       return pl;
     }
     uint8_t type = in.sig_script[siglen];
@@ -523,7 +548,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     if (!parse_pushes(nullptr, in.sig_script, in.sig_script_len, in.sig_script_off,
                       pushes) ||
         pushes.empty()) {
-      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
+      interp_classify(ctx, tx, in, input_index, sigop_units, pl);
       return pl;
     }
     /* last push = redeem candidate (need the actual bytes: offset into blob is
@@ -548,7 +573,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     pl.spent_units += 1; /* OpEqual pushed [1] */
     /* parse canonical multisig redeem: OP_m (0x20 pk)×n OP_n 0xae */
     if (rlen < 3 || redeem[0] < 0x51 || redeem[0] > 0x60) {
-      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
+      interp_classify(ctx, tx, in, input_index, sigop_units, pl);
       return pl;
     }
     int m = redeem[0] - 0x50;
@@ -561,7 +586,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     int n = (int)key_offs.size();
     if (rp + 2 != rlen || n < 1 || n > 20 || redeem[rp] != (uint8_t)(0x50 + n) ||
         redeem[rp + 1] != 0xae || m > n || m < 1) {
-      pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
+      interp_classify(ctx, tx, in, input_index, sigop_units, pl);
       return pl;
     }
     int nsigs = (int)pushes.size() - 1;
@@ -580,7 +605,20 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     pl.msig_n = n;
     pl.key_offs = key_offs;
     /* sigs = the last m pushes before the redeem (stack top-down ordering:
-     * multisig pops the TOP m entries = the last m pushes) */
+     * multisig pops the TOP m entries = the last m pushes).
+     * A sig whose length is neither 0 nor 65 needs the current KEY's parse
+     * status before its own length error (check order: cost, pubkey,
+     * signature — lib.rs:884-890), which the template fast path cannot
+     * provide; such inputs run through the general interpreter instead. */
+    for (int si = nsigs - m; si < nsigs; si++) {
+      uint32_t sl = pushes[si].second;
+      if (sl != 0 && sl != 65) {
+        pl = InputPlan();
+        pl.limit_units = committed_limit(in);
+        interp_classify(ctx, tx, in, input_index, sigop_units, pl);
+        return pl;
+      }
+    }
     for (int si = nsigs - m; si < nsigs; si++) {
       MsigSig ms;
       ms.off = pushes[si].first;
@@ -607,7 +645,7 @@ static InputPlan classify_input(const HTx &tx, const HInput &in,
     return pl;
   }
 
-  pl.pre_code = unsupported_pre_code(tx, in, input_index, sigop_units);
+  interp_classify(ctx, tx, in, input_index, sigop_units, pl);
   return pl;
 }
 
@@ -854,7 +892,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     plans[t].reserve(tx.inputs.size());
     for (uint32_t i = 0; i < tx.inputs.size(); i++)
       plans[t].push_back(
-          classify_input(tx, tx.inputs[i], sigop_units, sjobs, ejobs, t, i));
+          classify_input(ctx, tx, tx.inputs[i], sigop_units, sjobs, ejobs, t, i));
   };
   {
     if (P1T == 1) {
@@ -1053,6 +1091,180 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
   } else {
     s_status.swap(s_gpu);
     e_status.swap(e_gpu);
+  }
+
+  /* phase 2.5: interpreter rounds — resolve suspended general-interpreter
+   * scripts over the GPU verify batch, one signature site per script per
+   * round (collect/replay protocol, kv_script_host.inc). Rounds =
+   * max signature-site depth over this block's non-template scripts (almost
+   * always 1); each round is one GPU batch over ALL suspended scripts, so
+   * the launch count does not scale with signature count. */
+  if (flags != KV_FLAGS_SKIP_SCRIPT_CHECKS) {
+    std::vector<std::pair<uint32_t, uint32_t>> islots;
+    for (int t = 0; t < n_txs; t++) {
+      if (codes[t]) continue;
+      for (uint32_t i = 0; i < plans[t].size(); i++)
+        if (plans[t][i].kind == PLAN_INTERP) islots.emplace_back((uint32_t)t, i);
+    }
+    std::vector<uint8_t> st_h, et_h; /* host-side tuple arrays per round */
+    while (!islots.empty()) {
+      st_h.clear();
+      et_h.clear();
+      std::vector<kv::kv_job> mjobs_s, mjobs_e; /* sighash-msg jobs */
+      struct ReqRef {
+        uint8_t ecdsa;
+        uint32_t idx;
+      };
+      std::vector<std::vector<ReqRef>> refs(islots.size());
+      for (size_t sl = 0; sl < islots.size(); sl++) {
+        uint32_t t = islots[sl].first, i = islots[sl].second;
+        InputPlan &pl = plans[t][i];
+        const HInput &in = txs[t].inputs[i];
+        for (const auto &rq : pl.pending) {
+          if (!rq.ecdsa) {
+            uint32_t idx = (uint32_t)(st_h.size() / 128);
+            st_h.resize(st_h.size() + 128);
+            uint8_t *tp = st_h.data() + (size_t)idx * 128;
+            memcpy(tp, rq.sig, 64);
+            memcpy(tp + 64, rq.pk, 32);
+            if (rq.literal)
+              memcpy(tp + 96, rq.msg, 32);
+            else /* sig_off carries the tuple slot for the msg kernel */
+              mjobs_s.push_back(
+                  kv::kv_job{t, in.rec_off, i, idx, 0, rq.hash_type, 0, 0});
+            refs[sl].push_back({0, idx});
+          } else {
+            uint32_t idx = (uint32_t)(et_h.size() / 132);
+            et_h.resize(et_h.size() + 132);
+            uint8_t *tp = et_h.data() + (size_t)idx * 132;
+            memcpy(tp, rq.sig, 64);
+            memcpy(tp + 64, rq.pk, 33);
+            if (rq.literal)
+              memcpy(tp + 97, rq.msg, 32);
+            else
+              mjobs_e.push_back(
+                  kv::kv_job{t, in.rec_off, i, idx, 0, rq.hash_type, 1, 0});
+            refs[sl].push_back({1, idx});
+          }
+        }
+      }
+      size_t ns_i = st_h.size() / 128, ne_i = et_h.size() / 132;
+      std::vector<uint8_t> s_st(ns_i), e_st(ne_i);
+      if (ns_i + ne_i > 0) {
+        if (!blob_uploaded) {
+          if (g_vb.blob.ensure(blob_len) ||
+              g_vb.subhashes.ensure((size_t)n_txs * 160))
+            return -2;
+          HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len,
+                                   hipMemcpyHostToDevice, ctx->stream));
+          hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256),
+                             dim3(256), 0, ctx->stream,
+                             (const uint8_t *)g_vb.blob.p, (uint32_t)n_txs,
+                             (uint8_t *)g_vb.subhashes.p);
+          blob_uploaded = true;
+        }
+        if (ns_i) {
+          if (g_vb.s_tuples.ensure(st_h.size()) ||
+              g_vb.s_bitmap.ensure((ns_i + 63) / 64 * 8) ||
+              g_vb.s_status.ensure(ns_i) ||
+              (!mjobs_s.empty() &&
+               g_vb.s_jobs.ensure(mjobs_s.size() * sizeof(kv::kv_job))))
+            return -2;
+          HIP_CHECK(hipMemcpyAsync(g_vb.s_tuples.p, st_h.data(), st_h.size(),
+                                   hipMemcpyHostToDevice, ctx->stream));
+          if (!mjobs_s.empty()) {
+            HIP_CHECK(hipMemcpyAsync(g_vb.s_jobs.p, mjobs_s.data(),
+                                     mjobs_s.size() * sizeof(kv::kv_job),
+                                     hipMemcpyHostToDevice, ctx->stream));
+            hipLaunchKernelGGL(kv::kv_sighash_msg_kernel,
+                               dim3(((uint32_t)mjobs_s.size() + 255) / 256),
+                               dim3(256), 0, ctx->stream,
+                               (const uint8_t *)g_vb.blob.p,
+                               (const uint8_t *)g_vb.subhashes.p,
+                               (const kv::kv_job *)g_vb.s_jobs.p,
+                               (uint32_t)mjobs_s.size(),
+                               (uint8_t *)g_vb.s_tuples.p, 128u, 96u);
+          }
+          hipLaunchKernelGGL(kv::kv_schnorr_verify_kernel,
+                             dim3(((uint32_t)ns_i + 255) / 256), dim3(256), 0,
+                             ctx->stream, (const uint8_t *)g_vb.s_tuples.p,
+                             (unsigned long long)ns_i,
+                             (unsigned long long *)g_vb.s_bitmap.p,
+                             (uint8_t *)g_vb.s_status.p);
+          HIP_CHECK(hipMemcpyAsync(s_st.data(), g_vb.s_status.p, ns_i,
+                                   hipMemcpyDeviceToHost, ctx->stream));
+        }
+        if (ne_i) {
+          if (g_vb.e_tuples.ensure(et_h.size()) ||
+              g_vb.e_bitmap.ensure((ne_i + 63) / 64 * 8) ||
+              g_vb.e_status.ensure(ne_i) ||
+              (!mjobs_e.empty() &&
+               g_vb.e_jobs.ensure(mjobs_e.size() * sizeof(kv::kv_job))))
+            return -2;
+          HIP_CHECK(hipMemcpyAsync(g_vb.e_tuples.p, et_h.data(), et_h.size(),
+                                   hipMemcpyHostToDevice, ctx->stream));
+          if (!mjobs_e.empty()) {
+            HIP_CHECK(hipMemcpyAsync(g_vb.e_jobs.p, mjobs_e.data(),
+                                     mjobs_e.size() * sizeof(kv::kv_job),
+                                     hipMemcpyHostToDevice, ctx->stream));
+            hipLaunchKernelGGL(kv::kv_sighash_msg_kernel,
+                               dim3(((uint32_t)mjobs_e.size() + 255) / 256),
+                               dim3(256), 0, ctx->stream,
+                               (const uint8_t *)g_vb.blob.p,
+                               (const uint8_t *)g_vb.subhashes.p,
+                               (const kv::kv_job *)g_vb.e_jobs.p,
+                               (uint32_t)mjobs_e.size(),
+                               (uint8_t *)g_vb.e_tuples.p, 132u, 97u);
+          }
+          hipLaunchKernelGGL(kv::kv_ecdsa_verify_kernel,
+                             dim3(((uint32_t)ne_i + 255) / 256), dim3(256), 0,
+                             ctx->stream, (const uint8_t *)g_vb.e_tuples.p,
+                             (unsigned long long)ne_i,
+                             (unsigned long long *)g_vb.e_bitmap.p,
+                             (uint8_t *)g_vb.e_status.p);
+          HIP_CHECK(hipMemcpyAsync(e_st.data(), g_vb.e_status.p, ne_i,
+                                   hipMemcpyDeviceToHost, ctx->stream));
+        }
+        HIP_CHECK(hipGetLastError());
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+      }
+      /* feed statuses back as this site's memo entry and replay */
+      std::vector<uint8_t> survive(islots.size(), 0);
+      kvh_parallel_for((uint32_t)islots.size(), [&](uint32_t sl) {
+        uint32_t t = islots[sl].first, i = islots[sl].second;
+        InputPlan &pl = plans[t][i];
+        std::vector<uint8_t> entry(refs[sl].size());
+        for (size_t r = 0; r < refs[sl].size(); r++)
+          entry[r] = refs[sl][r].ecdsa ? e_st[refs[sl][r].idx]
+                                       : s_st[refs[sl][r].idx];
+        pl.memo.push_back(std::move(entry));
+        pl.pending.clear();
+        kvhost::KvsRunCtx rctx;
+        rctx.memo = &pl.memo;
+        rctx.pending = &pl.pending;
+        rctx.seqc_fn = ctx->seqc_fn;
+        rctx.seqc_user = ctx->seqc_user;
+        const HInput &in = txs[t].inputs[i];
+        int rc = kvh_run_input_script(txs[t], in, i, sigop_units, &rctx);
+        if (rc == kvhost::KVH_SCRIPT_SUSPEND) {
+          survive[sl] = 1;
+          return;
+        }
+        pl.kind = PLAN_NONE;
+        if (rc == kvhost::KVH_SCRIPT_DEFER)
+          pl.pre_code = KV_TX_DEFER_TO_CPU;
+        else if (rc != 0)
+          pl.pre_code = (in.sig_script_len == 0 ? KV_ERR_SIGNATURE_EMPTY_BASE
+                                                : KV_ERR_SIGNATURE_INVALID_BASE) +
+                        rc;
+        else
+          pl.pre_code = 0;
+      });
+      std::vector<std::pair<uint32_t, uint32_t>> next;
+      for (size_t sl = 0; sl < islots.size(); sl++)
+        if (survive[sl]) next.push_back(islots[sl]);
+      islots.swap(next);
+    }
   }
 
   /* phase 3: resolution (first failing input wins, sequential semantics);

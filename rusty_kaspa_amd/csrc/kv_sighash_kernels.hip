@@ -120,13 +120,12 @@ struct kv_job {
   uint16_t _pad;
 };
 
-extern "C" __global__ void kv_sighash_assemble_kernel(
-    const uint8_t *__restrict__ blob, const uint8_t *__restrict__ subhashes,
-    const kv_job *__restrict__ jobs, uint32_t n_jobs,
-    uint8_t *__restrict__ schnorr_tuples, uint8_t *__restrict__ ecdsa_tuples) {
-  uint32_t ji = blockIdx.x * blockDim.x + threadIdx.x;
-  if (ji >= n_jobs) return;
-  kv_job job = jobs[ji];
+/* Compute the 32B signing hash for one job (the shared core of the assemble
+ * and msg-only kernels) ⇔ calc_schnorr/ecdsa_signature_hash
+ * (sighash.rs:245-292). */
+__device__ static void kv_compute_sighash_msg(const uint8_t *__restrict__ blob,
+                                              const uint8_t *__restrict__ subhashes,
+                                              const kv_job &job, uint8_t msg[32]) {
   blob_tx tx;
   blob_tx_at(blob, job.tx_index, tx);
   blob_input in;
@@ -193,7 +192,6 @@ extern "C" __global__ void kv_sighash_assemble_kernel(
   b2b_update_u64(S, tx.gas);
   b2b_update(S, sub + 128, 32); /* payload hash */
   b2b_update(S, &ht, 1);
-  uint8_t msg[32];
   b2b_final(S, msg);
 
   if (job.ecdsa) {
@@ -218,6 +216,19 @@ extern "C" __global__ void kv_sighash_assemble_kernel(
       msg[4 * i + 2] = (uint8_t)(h[i] >> 8);
       msg[4 * i + 3] = (uint8_t)h[i];
     }
+  }
+}
+
+extern "C" __global__ void kv_sighash_assemble_kernel(
+    const uint8_t *__restrict__ blob, const uint8_t *__restrict__ subhashes,
+    const kv_job *__restrict__ jobs, uint32_t n_jobs,
+    uint8_t *__restrict__ schnorr_tuples, uint8_t *__restrict__ ecdsa_tuples) {
+  uint32_t ji = blockIdx.x * blockDim.x + threadIdx.x;
+  if (ji >= n_jobs) return;
+  kv_job job = jobs[ji];
+  uint8_t msg[32];
+  kv_compute_sighash_msg(blob, subhashes, job, msg);
+  if (job.ecdsa) {
     uint8_t *t = ecdsa_tuples + (size_t)ji * 132;
     for (int i = 0; i < 64; i++) t[i] = blob[job.sig_off + i];
     for (int i = 0; i < 33; i++) t[64 + i] = blob[job.pk_off + i];
@@ -229,6 +240,24 @@ extern "C" __global__ void kv_sighash_assemble_kernel(
     for (int i = 0; i < 32; i++) t[64 + i] = blob[job.pk_off + i];
     for (int i = 0; i < 32; i++) t[96 + i] = msg[i];
   }
+}
+
+/* msg-only variant for interpreter-collected verify requests: sig/pk come
+ * from the SCRIPT STACK (host-prefilled in the tuple buffer), only the
+ * signing hash is computed here. job.sig_off is repurposed as the TUPLE SLOT:
+ * the hash lands at out + sig_off*stride + msg_off (not every tuple needs a
+ * sighash — literal-msg requests have none, so slots are sparse in jobs). */
+extern "C" __global__ void kv_sighash_msg_kernel(
+    const uint8_t *__restrict__ blob, const uint8_t *__restrict__ subhashes,
+    const kv_job *__restrict__ jobs, uint32_t n_jobs, uint8_t *__restrict__ out,
+    uint32_t stride, uint32_t msg_off) {
+  uint32_t ji = blockIdx.x * blockDim.x + threadIdx.x;
+  if (ji >= n_jobs) return;
+  kv_job job = jobs[ji];
+  uint8_t msg[32];
+  kv_compute_sighash_msg(blob, subhashes, job, msg);
+  uint8_t *t = out + (size_t)job.sig_off * stride + msg_off;
+  for (int i = 0; i < 32; i++) t[i] = msg[i];
 }
 
 /* ---------------- MuHash ---------------- */

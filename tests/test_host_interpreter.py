@@ -1,7 +1,13 @@
 """The PRODUCT host interpreter (kv_script_host.inc — the non-template
 fallback inside classify_input) differentially tested against the oracle:
-every canonical script vector row, plus random-script fuzz. Rows reaching a
-signature opcode DEFER (KVH_SCRIPT_DEFER) by design — the GPU owns EC."""
+every canonical script vector row, plus random-script fuzz.
+
+Signature-family opcodes execute through the collect/replay protocol: the
+interpreter suspends at each signature site, this harness resolves the
+pending verify requests with the PRODUCT's own verify functions (the HIP
+device code host-compiled in tests/host_shim/main.cpp — the same code path
+the GPU runs) and sighashes from the oracle, then replays. Only
+OpZkPrecompile still defers."""
 import ctypes
 import json
 import os
@@ -18,12 +24,12 @@ from test_script_vectors import (BuildError, CODE_NAMES, GOLD, assemble,
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 SHIM_DIR = os.path.join(REPO, "tests", "host_shim")
 LIB = os.path.join(SHIM_DIR, "libscriptshim.so")
+VLIB = os.path.join(SHIM_DIR, "libhostshim.so")
 DEFER = -100
+SUSPEND = -101
 UNSUPPORTED = 63
-# opcodes whose oracle path is supported but the product defers (sig family,
-# blake3, seq-commit, zk)
-DEFER_OPS = {0xa6, 0xa9, 0xab, 0xac, 0xad, 0xae, 0xaf, 0xd4, 0xd7, 0xd8,
-             0xd9, 0xda}
+# opcodes the engine does not decide (routed to the reference CPU)
+DEFER_OPS = {0xa6}
 
 
 @pytest.fixture(scope="module")
@@ -40,6 +46,23 @@ def shim():
                         src, "-o", LIB], check=True)
     lib = ctypes.CDLL(LIB)
     lib.host_run_input_script.restype = ctypes.c_int
+    lib.host_run_input_script_collect.restype = ctypes.c_int
+    return lib
+
+
+@pytest.fixture(scope="module")
+def verifier():
+    """The PRODUCT's EC verify functions, host-compiled (main.cpp shim)."""
+    src = os.path.join(SHIM_DIR, "main.cpp")
+    if (not os.path.exists(VLIB)
+            or os.path.getmtime(VLIB) < os.path.getmtime(src)):
+        subprocess.run(["g++", "-O1", "-fPIC", "-shared",
+                        "-I", os.path.join(REPO, "rusty_kaspa_amd", "csrc"),
+                        "-I", SHIM_DIR, src, "-o", VLIB], check=True)
+    lib = ctypes.CDLL(VLIB)
+    lib.host_init_gtable()
+    lib.host_schnorr_verify.restype = ctypes.c_int
+    lib.host_ecdsa_verify.restype = ctypes.c_int
     return lib
 
 
@@ -48,10 +71,51 @@ def run_product(shim, blob):
                                       ctypes.c_uint64(1000))
 
 
-def test_vector_suite_differential(oracle, shim):
+def run_product_resolved(shim, verifier, oracle, blob, max_rounds=64):
+    """Full collect/replay loop: resolve each suspended signature site with
+    the product's verify functions + the oracle's sighash, then replay."""
+    memo = []  # list of bytes (status vector per site)
+    cap = 512
+    pend = (ctypes.c_uint8 * (136 * cap))()
+    npend = ctypes.c_uint32(0)
+    for _ in range(max_rounds):
+        flat = b"".join(memo)
+        sizes = (ctypes.c_uint32 * max(1, len(memo)))(*[len(m) for m in memo])
+        rc = shim.host_run_input_script_collect(
+            blob, ctypes.c_size_t(len(blob)), 0, 0, ctypes.c_uint64(1000),
+            flat, sizes, ctypes.c_uint32(len(memo)), pend, ctypes.c_uint32(cap),
+            ctypes.byref(npend))
+        if rc != SUSPEND:
+            return rc
+        statuses = bytearray()
+        for r in range(npend.value):
+            rec = bytes(pend[136 * r:136 * (r + 1)])
+            ecdsa, literal, ht = rec[0], rec[1], rec[2]
+            sig, pk = rec[4:68], rec[68:101]
+            if literal:
+                msg = rec[101:133]
+            else:
+                out = (ctypes.c_uint8 * 32)()
+                rcs = oracle.ok_sighash(blob, ctypes.c_size_t(len(blob)), 0, 0,
+                                        ht, ecdsa, out)
+                assert rcs == 0
+                msg = bytes(out)
+            if ecdsa:
+                v = verifier.host_ecdsa_verify(pk, msg, sig)
+                st = {1: 0, 0: 1, -1: 2, -2: 3}[v]
+            else:
+                v = verifier.host_schnorr_verify(pk[:32], msg, sig)
+                st = {1: 0, 0: 1, -1: 2}[v]
+            statuses.append(st)
+        memo.append(bytes(statuses))
+    raise AssertionError("collect/replay did not converge")
+
+
+def test_vector_suite_differential(oracle, shim, verifier):
     blk = b"input_block".ljust(32, b"f")
     com = b"output_root_hash".ljust(32, b"f")
     oracle.ok_script_set_seq_commit_mock(blk, com)
+    shim.host_set_seq_commit_mock(blk, com)
     rows = json.load(open(os.path.join(GOLD, "script_tests.json")))
     ran = deferred = 0
     failures = []
@@ -66,7 +130,7 @@ def test_vector_suite_differential(oracle, shim):
         blob = build_case_blob(oracle, sig, spk)
         ocode = oracle.ok_check_input_script(blob, ctypes.c_size_t(len(blob)),
                                              0, 0)
-        pcode = run_product(shim, blob)
+        pcode = run_product_resolved(shim, verifier, oracle, blob)
         if pcode == DEFER:
             deferred += 1
             continue
@@ -75,14 +139,15 @@ def test_vector_suite_differential(oracle, shim):
         else:
             ran += 1
     print(f"\n[product interpreter] {ran} rows identical to the oracle, "
-          f"{deferred} deferred (signature family), {len(failures)} diverged")
+          f"{deferred} deferred (zk precompile only), {len(failures)} diverged")
     assert not failures, failures[:10]
-    assert ran > 700
+    assert ran > 940  # the full suite minus non-executable rows
+    assert deferred <= 5
 
 
-def test_random_script_fuzz(oracle, shim):
+def test_random_script_fuzz(oracle, shim, verifier):
     """Random byte scripts: the product interpreter and the oracle must agree
-    on every outcome (or the product defers on a sig opcode)."""
+    on every outcome (the sig family now executes; only zk defers)."""
     rng = random.Random(99)
     agreed = deferred = 0
     mismatches = []
@@ -99,7 +164,7 @@ def test_random_script_fuzz(oracle, shim):
         blob = build_case_blob(oracle, bytes(sig), spk)
         ocode = oracle.ok_check_input_script(blob, ctypes.c_size_t(len(blob)),
                                              0, 0)
-        pcode = run_product(shim, blob)
+        pcode = run_product_resolved(shim, verifier, oracle, blob)
         if pcode == DEFER:
             deferred += 1
             continue
@@ -110,3 +175,114 @@ def test_random_script_fuzz(oracle, shim):
     print(f"\n[fuzz] {agreed} agreed, {deferred} deferred, "
           f"{len(mismatches)} mismatched")
     assert not mismatches, mismatches[:5]
+
+
+def test_sig_family_structured_fuzz(oracle, shim, verifier):
+    """Structured fuzz aimed at the signature family: random checksig /
+    multisig / fromstack scripts with real, corrupted, truncated and empty
+    signatures, sig-dependent conditionals included."""
+    rng = random.Random(1234)
+    agreed = 0
+    mismatches = []
+
+    def push(data):
+        assert len(data) <= 0x4b
+        return bytes([len(data)]) + data if data else b"\x00"
+
+    for trial in range(1500):
+        mode = rng.randrange(6)
+        key = bytearray(rng.randrange(256) for _ in range(32))
+        sig64 = bytearray(rng.randrange(256) for _ in range(64))
+        ht = rng.choice([0x01, 0x02, 0x04, 0x81, 0x82, 0x84, 0x00, 0x33])
+        sigpush = bytes(sig64) + bytes([ht])
+        if mode == 0:  # bare checksig (schnorr/ecdsa)
+            op = rng.choice([0xac, 0xab])
+            k = key if op == 0xac else bytes([rng.choice([2, 3, 4])]) + bytes(key)
+            spk = push(bytes(k)) + bytes([op])
+            sig_script = push(sigpush)
+        elif mode == 1:  # checksig-verify + trailing true
+            spk = push(bytes(key)) + b"\xad\x51"
+            sig_script = push(sigpush)
+        elif mode == 2:  # multisig m-of-n with random sigs
+            n = rng.randrange(1, 4)
+            m = rng.randrange(0, n + 1)
+            keys = [bytes(rng.randrange(256) for _ in range(32))
+                    for _ in range(n)]
+            sigs = []
+            for _ in range(m):
+                if rng.randrange(4) == 0:
+                    sigs.append(b"")  # empty sig
+                elif rng.randrange(5) == 0:
+                    sigs.append(bytes(rng.randrange(256)
+                                      for _ in range(rng.randrange(1, 70))))
+                else:
+                    sigs.append(bytes(rng.randrange(256) for _ in range(64)) +
+                                bytes([rng.choice([1, 1, 1, 0x33])]))
+            spk = (bytes([0x50 + m]) if m else b"\x00")
+            spk = b"".join(push(k) for k in keys)
+            redeem = (bytes([0x50 + m if m else 0x00]) +
+                      b"".join(push(k) for k in keys) +
+                      bytes([0x50 + n, rng.choice([0xae, 0xa9, 0xaf])]))
+            # run redeem directly as spk (no p2sh wrapper needed for parity)
+            spk = redeem
+            sig_script = b"".join(push(s) for s in sigs)
+        elif mode == 3:  # checksig-from-stack with random msg
+            msg = bytes(rng.randrange(256)
+                        for _ in range(rng.choice([32, 32, 31, 33, 0])))
+            op = rng.choice([0xd7, 0xd8])
+            k = key if op == 0xd7 else bytes([rng.choice([2, 3])]) + bytes(key)
+            spk = push(bytes(sig64)) + push(msg) + push(bytes(k)) + bytes([op])
+            sig_script = b""
+        elif mode == 4:  # verdict-dependent conditional
+            spk = (push(bytes(key)) + b"\xac" +          # checksig → bool
+                   b"\x63" + b"\x51" + b"\x67" + b"\x52" + b"\x68" +  # if 1 else 2 endif
+                   bytes([0x51 + rng.randrange(2), 0x87]))  # eq check
+            sig_script = push(sigpush)
+        else:  # double checksig (two sites, sequential rounds)
+            spk = (push(bytes(key)) + b"\xac" +
+                   push(bytes(key)) + b"\xac" + b"\x9a")  # booland
+            sig_script = push(sigpush) + push(sigpush)
+        try:
+            blob = build_case_blob(oracle, sig_script, spk)
+        except Exception:
+            continue
+        ocode = oracle.ok_check_input_script(blob, ctypes.c_size_t(len(blob)),
+                                             0, 0)
+        pcode = run_product_resolved(shim, verifier, oracle, blob)
+        if pcode != ocode:
+            mismatches.append((mode, sig_script.hex()[:60], spk.hex()[:80],
+                               ocode, pcode))
+        else:
+            agreed += 1
+    print(f"\n[sig-fuzz] {agreed} agreed, {len(mismatches)} mismatched")
+    assert not mismatches, mismatches[:5]
+
+
+def test_real_signature_through_interpreter(oracle, shim, verifier):
+    """A REAL signature over the mock tx context must validate end-to-end
+    through the interpreter's collect/replay (not the template path):
+    spk = <pk> OP_CHECKSIG OP_VERIFY OP_TRUE so the shape is non-template."""
+    # build the blob with a placeholder to learn the sighash, then sign
+    key = bytes([7]) * 31 + b"\x01"
+    pk = (ctypes.c_uint8 * 32)()
+    assert oracle.ok_pubkey_xonly(key, pk) == 1
+    spk = bytes([0x20]) + bytes(pk) + b"\xac"  # P2PK shape but run via shim
+    blob = build_case_blob(oracle, b"\x41" + b"\x00" * 65, spk)
+    msg = (ctypes.c_uint8 * 32)()
+    assert oracle.ok_sighash(blob, ctypes.c_size_t(len(blob)), 0, 0, 1, 0,
+                             msg) == 0
+    sig = (ctypes.c_uint8 * 64)()
+    assert oracle.ok_schnorr_sign(key, msg, None, sig) == 1
+    good = b"\x41" + bytes(sig) + b"\x01"
+    blob = build_case_blob(oracle, good, spk)
+    ocode = oracle.ok_check_input_script(blob, ctypes.c_size_t(len(blob)), 0, 0)
+    pcode = run_product_resolved(shim, verifier, oracle, blob)
+    assert ocode == 0, ocode  # sanity: the oracle accepts the real signature
+    assert pcode == 0, pcode
+    # corrupt one signature byte → both reject identically (EvalFalse)
+    bad = bytearray(good)
+    bad[10] ^= 1
+    blob = build_case_blob(oracle, bytes(bad), spk)
+    ocode = oracle.ok_check_input_script(blob, ctypes.c_size_t(len(blob)), 0, 0)
+    pcode = run_product_resolved(shim, verifier, oracle, blob)
+    assert pcode == ocode == 1, (ocode, pcode)  # EvalFalse
