@@ -277,7 +277,10 @@ static int check_schnorr(eng *E, uint8_t hash_type, const sent *key, const uint8
                          int *valid) {
   int rc = consume_sigop(E);
   if (rc) return rc;
-  if (key->len != 32) return KV_SCRIPT_INVALID_PUBKEY; /* XOnly from_slice len */
+  /* XOnlyPublicKey::from_slice parses FULLY (length AND curve) before the
+   * signature is looked at (lib.rs:857-859) */
+  if (key->len != 32) return KV_SCRIPT_INVALID_PUBKEY;
+  if (!ok_xonly_pubkey_valid(key->d)) return KV_SCRIPT_INVALID_PUBKEY;
   if (sig_len != 64) return KV_SCRIPT_INVALID_SIGNATURE;
   uint8_t msg[32];
   if (from_stack) {
@@ -296,7 +299,10 @@ static int check_ecdsa(eng *E, uint8_t hash_type, const sent *key, const uint8_t
                        int *valid) {
   int rc = consume_sigop(E);
   if (rc) return rc;
+  /* check_pub_key_encoding_ecdsa (length), then PublicKey::from_slice
+   * (prefix + curve) — both before the signature (lib.rs:888-890) */
   if (key->len != 33) return KV_SCRIPT_PUBKEY_FORMAT;
+  if (!ok_compressed_pubkey_valid(key->d)) return KV_SCRIPT_INVALID_PUBKEY;
   if (sig_len != 64) return KV_SCRIPT_INVALID_SIGNATURE;
   uint8_t msg[32];
   if (from_stack) {

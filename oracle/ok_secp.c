@@ -648,6 +648,21 @@ static void tagged_hash(const char *tag, const uint8_t *d1, size_t l1,
   ok_sha256_final(&S, out32);
 }
 
+
+/* pubkey parse-only checks (XOnlyPublicKey::from_slice /
+ * PublicKey::from_slice succeed): 1 = parseable, 0 = not on curve / bad
+ * prefix / x >= p. Used by the script engine to order InvalidPubkey before
+ * signature-length errors, matching the reference's parse order. */
+int ok_xonly_pubkey_valid(const uint8_t pk32[32]) {
+  ge P;
+  return ge_lift_x_even(&P, pk32) ? 1 : 0;
+}
+
+int ok_compressed_pubkey_valid(const uint8_t pk33[33]) {
+  ge P;
+  return ge_parse_compressed(&P, pk33) ? 1 : 0;
+}
+
 int ok_schnorr_verify(const uint8_t pk32[32], const uint8_t msg32[32],
                       const uint8_t sig64[64]) {
   ge P;

@@ -190,3 +190,7 @@ int ok_test_storage_mass(uint32_t n_ins, const uint64_t *in_amounts,
 uint64_t ok_normalized_max_limits(uint64_t storage_mass, uint64_t compute_mass,
                                   uint64_t transient_mass, uint64_t limit_storage,
                                   uint64_t limit_compute, uint64_t limit_transient);
+
+/* pubkey parse-only validity (XOnlyPublicKey/PublicKey::from_slice succeed) */
+int ok_xonly_pubkey_valid(const uint8_t pk32[32]);
+int ok_compressed_pubkey_valid(const uint8_t pk33[33]);
