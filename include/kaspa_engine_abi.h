@@ -279,6 +279,22 @@ typedef int (*kv_seq_commit_accessor_fn)(void *user, const uint8_t block_hash32[
                                          uint8_t commitment_out32[32]);
 int kv_set_seq_commit_accessor(kv_ctx *ctx, kv_seq_commit_accessor_fn fn, void *user);
 
+/* Per-kernel GPU timings (hipEvents on the engine stream) and job counts of
+ * the LAST kv_validate_block/kv_validate_block_utxo call on this context —
+ * the bench harness derives the block-path kernel roofline from these.
+ * A kernel that did not launch reports 0. */
+typedef struct {
+  double subhash_ms;   /* kv_tx_subhash_kernel */
+  double s_assemble_ms; /* sighash+tuple assembly, schnorr jobs */
+  double e_assemble_ms; /* sighash+tuple assembly, ecdsa jobs */
+  double schnorr_ms;   /* kv_schnorr_verify_kernel */
+  double ecdsa_ms;     /* kv_ecdsa_verify_kernel */
+  double muhash_ms;    /* element + reduce kernels */
+  uint64_t n_schnorr;  /* schnorr verify lane-jobs (cache misses only) */
+  uint64_t n_ecdsa;    /* ecdsa verify lane-jobs */
+} kv_validate_timings;
+int kv_get_validate_timings(kv_ctx *ctx, kv_validate_timings *out);
+
 /* Sig-cache statistics (crypto/txscript/src/caches.rs:57-82 counters). */
 typedef struct {
   uint64_t insertions;

@@ -62,6 +62,11 @@ struct kv_ctx {
   /* KIP-21 seq-commitment accessor (kv_set_seq_commit_accessor) */
   kv_seq_commit_accessor_fn seqc_fn = nullptr;
   void *seqc_user = nullptr;
+  /* per-kernel timing events of the last validate call (kv_get_validate_timings):
+   * pairs (start,stop) for subhash, s-assemble, e-assemble, schnorr, ecdsa, muhash */
+  hipEvent_t tev[12] = {};
+  bool tev_init = false;
+  kv_validate_timings last_timings = {};
   /* grow-on-demand device scratch */
   uint8_t *d_in = nullptr;
   size_t d_in_cap = 0;
@@ -338,6 +343,13 @@ extern "C" int kv_set_seq_commit_accessor(kv_ctx *ctx, kv_seq_commit_accessor_fn
   std::lock_guard<std::mutex> lk(ctx->mu);
   ctx->seqc_fn = fn;
   ctx->seqc_user = user;
+  return 0;
+}
+
+extern "C" int kv_get_validate_timings(kv_ctx *ctx, kv_validate_timings *out) {
+  if (!ctx || !out) return -1;
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  *out = ctx->last_timings;
   return 0;
 }
 
@@ -802,6 +814,23 @@ extern "C" int kv_sighash_batch(kv_ctx *ctx, const uint8_t *blob, size_t blob_le
   return 0;
 }
 
+/* timing-event helpers (kv_get_validate_timings) */
+static void tev_ensure(kv_ctx *ctx) {
+  if (ctx->tev_init) return;
+  for (int i = 0; i < 12; i++) (void)hipEventCreate(&ctx->tev[i]);
+  ctx->tev_init = true;
+}
+static inline void tev_rec(kv_ctx *ctx, int i) {
+  (void)hipEventRecord(ctx->tev[i], ctx->stream);
+}
+static double tev_ms(kv_ctx *ctx, int pair) {
+  float ms = 0.f;
+  if (hipEventElapsedTime(&ms, ctx->tev[2 * pair], ctx->tev[2 * pair + 1]) !=
+      hipSuccess)
+    return 0.0;
+  return (double)ms;
+}
+
 /* caller holds ctx->mu; pre_codes (optional) pre-fails txs (e.g. missing
  * outpoints from the populate step) so they skip validation and muhash */
 static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len,
@@ -816,6 +845,9 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     return -1;
   }
   uint64_t sigop_units = ctx->params.mass_per_sig_op * KVH_UNITS_PER_GRAM;
+  tev_ensure(ctx);
+  bool tev_rec_pair[6] = {false, false, false, false, false, false};
+  ctx->last_timings = kv_validate_timings{};
 
   /* phase 1: host integer checks + classification, fanned over the host cores
    * (⇔ the reference's rayon pool) with per-chunk job lists so the GPU job
@@ -1026,9 +1058,12 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
                              ctx->stream));
     blob_uploaded = true;
+    tev_rec(ctx, 0);
     hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256), dim3(256),
                        0, ctx->stream, (const uint8_t *)g_vb.blob.p, (uint32_t)n_txs,
                        (uint8_t *)g_vb.subhashes.p);
+    tev_rec(ctx, 1);
+    tev_rec_pair[0] = true;
     if (ns) {
       if (g_vb.s_jobs.ensure(ns * sizeof(kv::kv_job)) ||
           g_vb.s_tuples.ensure(ns * 128) || g_vb.s_bitmap.ensure((ns + 63) / 64 * 8) ||
@@ -1036,15 +1071,22 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
         return -2;
       HIP_CHECK(hipMemcpyAsync(g_vb.s_jobs.p, sjobs.data(), ns * sizeof(kv::kv_job),
                                hipMemcpyHostToDevice, ctx->stream));
+      tev_rec(ctx, 2);
       hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                          dim3(((uint32_t)ns + 255) / 256), dim3(256), 0, ctx->stream,
                          (const uint8_t *)g_vb.blob.p, (const uint8_t *)g_vb.subhashes.p,
                          (const kv::kv_job *)g_vb.s_jobs.p, (uint32_t)ns,
                          (uint8_t *)g_vb.s_tuples.p, (uint8_t *)g_vb.s_tuples.p);
+      tev_rec(ctx, 3);
+      tev_rec_pair[1] = true;
+      tev_rec(ctx, 6);
       hipLaunchKernelGGL(kv::kv_schnorr_verify_kernel, dim3(((uint32_t)ns + 255) / 256),
                          dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.s_tuples.p,
                          (unsigned long long)ns, (unsigned long long *)g_vb.s_bitmap.p,
                          (uint8_t *)g_vb.s_status.p);
+      tev_rec(ctx, 7);
+      tev_rec_pair[3] = true;
+      ctx->last_timings.n_schnorr += ns;
       HIP_CHECK(hipMemcpyAsync(s_gpu.data(), g_vb.s_status.p, ns,
                                hipMemcpyDeviceToHost, ctx->stream));
     }
@@ -1055,15 +1097,22 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
         return -2;
       HIP_CHECK(hipMemcpyAsync(g_vb.e_jobs.p, ejobs.data(), ne * sizeof(kv::kv_job),
                                hipMemcpyHostToDevice, ctx->stream));
+      tev_rec(ctx, 4);
       hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                          dim3(((uint32_t)ne + 255) / 256), dim3(256), 0, ctx->stream,
                          (const uint8_t *)g_vb.blob.p, (const uint8_t *)g_vb.subhashes.p,
                          (const kv::kv_job *)g_vb.e_jobs.p, (uint32_t)ne,
                          (uint8_t *)g_vb.e_tuples.p, (uint8_t *)g_vb.e_tuples.p);
+      tev_rec(ctx, 5);
+      tev_rec_pair[2] = true;
+      tev_rec(ctx, 8);
       hipLaunchKernelGGL(kv::kv_ecdsa_verify_kernel, dim3(((uint32_t)ne + 255) / 256),
                          dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.e_tuples.p,
                          (unsigned long long)ne, (unsigned long long *)g_vb.e_bitmap.p,
                          (uint8_t *)g_vb.e_status.p);
+      tev_rec(ctx, 9);
+      tev_rec_pair[4] = true;
+      ctx->last_timings.n_ecdsa += ne;
       HIP_CHECK(hipMemcpyAsync(e_gpu.data(), g_vb.e_status.p, ne,
                                hipMemcpyDeviceToHost, ctx->stream));
     }
@@ -1320,6 +1369,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
       HIP_CHECK(hipMemcpyAsync(g_vb.elem_jobs.p, all.data(),
                                n_all * sizeof(kv::kv_elem_job), hipMemcpyHostToDevice,
                                ctx->stream));
+      tev_rec(ctx, 10);
       hipLaunchKernelGGL(kv::kv_muhash_element_kernel,
                          dim3(((uint32_t)n_all + 255) / 256), dim3(256), 0, ctx->stream,
                          (const uint8_t *)g_vb.blob.p,
@@ -1357,8 +1407,19 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
                                  ctx->stream));
         HIP_CHECK(hipStreamSynchronize(ctx->stream));
       }
+      tev_rec(ctx, 11);
+      tev_rec_pair[5] = true;
     }
   }
+
+  /* collect per-kernel timings (the stream is synchronized by now) */
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  if (tev_rec_pair[0]) ctx->last_timings.subhash_ms = tev_ms(ctx, 0);
+  if (tev_rec_pair[1]) ctx->last_timings.s_assemble_ms = tev_ms(ctx, 1);
+  if (tev_rec_pair[2]) ctx->last_timings.e_assemble_ms = tev_ms(ctx, 2);
+  if (tev_rec_pair[3]) ctx->last_timings.schnorr_ms = tev_ms(ctx, 3);
+  if (tev_rec_pair[4]) ctx->last_timings.ecdsa_ms = tev_ms(ctx, 4);
+  if (tev_rec_pair[5]) ctx->last_timings.muhash_ms = tev_ms(ctx, 5);
 
   for (int t = 0; t < n_txs; t++)
     if (codes[t]) fees[t] = 0; /* fee defined only for accepted txs */
