@@ -123,9 +123,9 @@ struct kv_job {
 /* Compute the 32B signing hash for one job (the shared core of the assemble
  * and msg-only kernels) ⇔ calc_schnorr/ecdsa_signature_hash
  * (sighash.rs:245-292). */
-__device__ static void kv_compute_sighash_msg(const uint8_t *__restrict__ blob,
-                                              const uint8_t *__restrict__ subhashes,
-                                              const kv_job &job, uint8_t msg[32]) {
+__device__ __forceinline__ static void kv_compute_sighash_msg(
+    const uint8_t *__restrict__ blob, const uint8_t *__restrict__ subhashes,
+    const kv_job job, uint8_t msg[32]) {
   blob_tx tx;
   blob_tx_at(blob, job.tx_index, tx);
   blob_input in;
