@@ -44,20 +44,22 @@ INVALID_PERMILLE = 0  # all-valid variant is the headline; 10%-invalid via flag
 DATA_DIR = os.path.join(REPO, "bench_data")
 
 # Algorithmic work accounting for the verify-kernel roofline (single committed
-# account, mirrored by DESIGN.md §4): the GLV-split 4-bit windowed ladder does
-#   132 Jacobian doubles x 7 fe_mul-equiv        =  924
-#    66 mixed adds       x 11                    =  726
-#    66 full adds        x 16                    = 1056
-#    66 phi-multiplies   x 1                     =   66
+# account, mirrored by DESIGN.md §4): the GLV-split 4-bit windowed ladder with
+# the AFFINE batch-inverted P table does
+#   132 Jacobian doubles  x  7 fe_mul-equiv      =  924
+#   132 mixed adds        x 11                   = 1452
+#    66 phi-multiplies    x  1                   =   66
 #    15-entry P-table build 14 x 11              =  154
+#    table batch-inversion (prefix 13 + inv 269
+#      + back-sub 26 + 4x14 per-entry)           =  364
 #    x-lift sqrt (addition chain)                =  266
 #    final inversion (addition chain)            =  269  (+ ~100 misc: scalar
-#    decomposition muls, challenge/normalize)    ≈ 3,170 fe_mul-equivalents
+#    decomposition muls, challenge/normalize)    ≈ 3,595 fe_mul-equivalents
 # per verify; each 10x26-limb fe_mul ≈ 170 u32-ALU-op equivalents (100 v_mad
-# 32x32 column products + fold/normalize) → ≈ 0.54e6 u32-ops per verify.
-ALG_FE_MULS_PER_VERIFY = 3170
+# 32x32 column products + fold/normalize) → ≈ 0.61e6 u32-ops per verify.
+ALG_FE_MULS_PER_VERIFY = 3595
 ALG_OPS_PER_FE_MUL = 170
-ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 538,900
+ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 611,150
 # gfx950 VALU issue peak: 256 CU x 4 SIMD x 32 lanes x 2.4 GHz = 78.6 T u32/s
 VALU_PEAK_TOPS = 78.6
 # HBM traffic per verify, measured by rocprofv3 --pmc (FETCH_SIZE+WRITE_SIZE,
@@ -337,7 +339,8 @@ def verify_mode(args):
 
     kernel_ms = ctypes.c_double()
     words = (n + 63) // 64
-    bitmap_t = torch.zeros(words, dtype=torch.int64, device=f"cuda:{local_rank}")
+    bitmap_t = (torch.zeros(words, dtype=torch.int64, device=f"cuda:{local_rank}")
+                if distributed else None)
 
     def one_step():
         rc = lib.kv_verify_staged(ctx, ctypes.c_size_t(n), 0, ctypes.byref(kernel_ms))
@@ -455,8 +458,13 @@ def main():
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    # torch must own HIP runtime initialization BEFORE the engine loads it:
+    # initializing the runtime through libkaspa_gpu first leaves torch's
+    # device enumeration empty ("No HIP GPUs are available")
+    import torch
+    if torch.cuda.is_available():
+        torch.cuda.init()
     if world > 1:
-        import torch
         import torch.distributed as dist
         dist.init_process_group(backend="nccl")
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))

@@ -254,14 +254,55 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
   glv_half g1h, g2h, p1h, p2h;
   glv_split(gs, g1h, g2h);
   glv_split(ps, p1h, p2h);
-  /* per-lane P table (Jacobian): ptab[k] = k·P, k = 1..15 */
-  gej ptab[16];
-  ptab[1].x = P.x;
-  ptab[1].y = P.y;
-  fe26_set_int(ptab[1].z, 1);
+  /* Per-lane P table in AFFINE form, batch-inverted with Montgomery's trick.
+   * The round-1 Jacobian table kept ~252MB of in-flight per-lane tables
+   * (16 entries x 120B x 131k resident lanes) — right at the 256MB MALL
+   * capacity — and its dynamic-indexed scratch loads paid ~44KB of fetch
+   * per verify (r01 PMC). Affine entries are 80B (working set ~168MB, fits
+   * MALL), and the P streams become MIXED adds (11 vs 16 fe_muls); the one
+   * extra field inversion (+~360 fe_muls with the back-substitution) is
+   * paid back by the 66 cheaper ladder adds (-330). */
+  ge ptab[16];
+  {
+    fe26 ztab[16], pref[16];
+    gej acc;
+    acc.x = P.x;
+    acc.y = P.y;
+    fe26_set_int(acc.z, 1);
+    ptab[1] = P;
 #pragma unroll 1
-  for (int k = 2; k <= 15; k++) gej_add_ge(ptab[k], ptab[k - 1], P);
-  gej_set_infinity(ptab[0]);
+    for (int k = 2; k <= 15; k++) {
+      gej t;
+      gej_add_ge(t, acc, P);
+      acc = t;
+      ptab[k].x = acc.x;
+      ptab[k].y = acc.y;
+      ztab[k] = acc.z;
+    }
+    /* batch-invert z2..z15: one fe26_inv + 3 muls per entry */
+    pref[2] = ztab[2];
+#pragma unroll 1
+    for (int k = 3; k <= 15; k++) fe26_mul(pref[k], pref[k - 1], ztab[k]);
+    fe26 inv;
+    fe26_inv(inv, pref[15]);
+#pragma unroll 1
+    for (int k = 15; k >= 2; k--) {
+      fe26 zi;
+      if (k > 2) {
+        fe26_mul(zi, inv, pref[k - 1]);
+        fe26_mul(inv, inv, ztab[k]);
+      } else {
+        zi = inv;
+      }
+      fe26 zi2, zi3;
+      fe26_sqr(zi2, zi);
+      fe26_mul(zi3, zi2, zi);
+      fe26_mul(ptab[k].x, ptab[k].x, zi2);
+      fe26_mul(ptab[k].y, ptab[k].y, zi3);
+    }
+    ptab[0] = ptab[1]; /* digit 0: add computed then discarded by cmov — a
+                          well-formed point keeps magnitudes in range */
+  }
   fe26 beta;
   {
     fe bu;
@@ -297,27 +338,27 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
       gej_add_ge(t, R, e);
       gej_cmov(R, t, (u64)(d != 0));
     }
-    /* P stream */
+    /* P stream (mixed add vs the affine table) */
     {
       u64 d = glv_digit(p1h, w);
-      gej e = ptab[d];
+      ge e = ptab[d];
       fe26 ny;
       fe26_neg(ny, e.y, 2);
       fe26_cmov(e.y, ny, (u32)p1h.neg);
-      gej_add(t, R, e);
+      gej_add_ge(t, R, e);
       gej_cmov(R, t, (u64)(d != 0));
     }
-    /* φP stream: (β·X, ±Y, Z) */
+    /* φP stream: (β·x, ±y) */
     {
       u64 d = glv_digit(p2h, w);
-      gej e = ptab[d];
+      ge e = ptab[d];
       fe26 bx;
       fe26_mul(bx, e.x, beta);
       e.x = bx;
       fe26 ny;
       fe26_neg(ny, e.y, 2);
       fe26_cmov(e.y, ny, (u32)p2h.neg);
-      gej_add(t, R, e);
+      gej_add_ge(t, R, e);
       gej_cmov(R, t, (u64)(d != 0));
     }
   }

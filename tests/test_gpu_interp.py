@@ -154,10 +154,11 @@ def test_checksig_from_stack(oracle, engine):
 
 
 def test_two_sites_two_rounds(oracle, engine):
-    """A script with two sequential checksig sites (two GPU rounds) and a
-    second site only reachable through the first verdict."""
+    """A script with two sequential checksig sites (two GPU rounds): the
+    first verdict parks on the alt stack while the second verifies."""
     key, pk = keypair(oracle, 13)
-    spk = push(pk) + b"\xac" + push(pk) + b"\xac\x9a"  # booland of two checksigs
+    # [s1 s2] pk CHECKSIG TOALT pk CHECKSIG FROMALT BOOLAND
+    spk = (push(pk) + b"\xac\x6b" + push(pk) + b"\xac\x6c\x9a")
     placeholder = push(bytes(65)) + push(bytes(65))
     txs = [spend_tx(oracle, placeholder, spk)]
     blob = B.build_blob(txs)
