@@ -6,7 +6,7 @@ O = ctypes.CDLL("oracle/liboracle.so")
 from rusty_kaspa_amd.engine import Engine
 n = 262144
 buf = ctypes.create_string_buffer(n*128)
-O.ok_gen_schnorr_tuples(ctypes.c_uint64(1), ctypes.c_size_t(n), 0, buf, 16)
+O.ok_gen_schnorr_tuples(ctypes.c_uint64(1), ctypes.c_size_t(n), 0, buf, os.cpu_count() or 16)
 eng = Engine(); lib = eng.lib; ctx = ctypes.c_void_p(eng.ctx)
 lib.kv_stage_tuples(ctx, buf, ctypes.c_size_t(n), 0)
 ms = ctypes.c_double()
