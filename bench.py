@@ -62,10 +62,12 @@ ALG_OPS_PER_FE_MUL = 170
 ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 611,150
 # gfx950 VALU issue peak: 256 CU x 4 SIMD x 32 lanes x 2.4 GHz = 78.6 T u32/s
 VALU_PEAK_TOPS = 78.6
-# HBM traffic per verify, measured by rocprofv3 --pmc (FETCH_SIZE+WRITE_SIZE,
-# corrected per MI355X_MICROARCH.md) on the staged 1M-tuple dispatch; see
-# profiles/ (r02 PMC run). None until a committed profile backs the number.
-TRAFFIC_BYTES_PER_VERIFY = None
+# Memory-side traffic per verify, measured by rocprofv3 --pmc FETCH_SIZE /
+# WRITE_SIZE (separate passes) on 262144-tuple staged dispatches:
+# (1.352e7 + 1.702e7) KiB / 262144 = 119,296 B/verify (fetch 52.8KB + write
+# 66.5KB; dominated by the __noinline__ group-op scratch ABI, vs 128B of
+# algorithmic input). Provenance: profiles/r02_rocprof_summary.json.
+TRAFFIC_BYTES_PER_VERIFY = 119296
 
 
 def log(msg):
