@@ -58,6 +58,10 @@ struct kv_sig_key_hash {
 struct kv_ctx {
   kv_params params;
   hipStream_t stream;
+  hipStream_t stream2; /* ecdsa verify chain (overlaps the schnorr chain) */
+  hipStream_t stream3; /* optimistic muhash chain (overlaps both) */
+  hipEvent_t ev_blob;  /* blob upload complete (sync-only, no timing) */
+  hipEvent_t ev_sub;   /* subhash kernel complete */
   std::mutex mu;
   /* KIP-21 seq-commitment accessor (kv_set_seq_commit_accessor) */
   kv_seq_commit_accessor_fn seqc_fn = nullptr;
@@ -124,8 +128,12 @@ extern "C" kv_ctx *kv_create(const kv_params *params) {
   if (params) ctx->params = *params;
   else
     ctx->params = kv_params{1000, 1000, 10000, 0};
-  if (hipStreamCreate(&ctx->stream) != hipSuccess) {
-    set_error("kv_create: hipStreamCreate failed");
+  if (hipStreamCreate(&ctx->stream) != hipSuccess ||
+      hipStreamCreate(&ctx->stream2) != hipSuccess ||
+      hipStreamCreate(&ctx->stream3) != hipSuccess ||
+      hipEventCreateWithFlags(&ctx->ev_blob, hipEventDisableTiming) != hipSuccess ||
+      hipEventCreateWithFlags(&ctx->ev_sub, hipEventDisableTiming) != hipSuccess) {
+    set_error("kv_create: stream/event creation failed");
     delete ctx;
     return nullptr;
   }
@@ -145,6 +153,10 @@ extern "C" void kv_destroy(kv_ctx *ctx) {
   if (ctx->d_bitmap) (void)hipFree(ctx->d_bitmap);
   if (ctx->d_status) (void)hipFree(ctx->d_status);
   (void)hipStreamDestroy(ctx->stream);
+  (void)hipStreamDestroy(ctx->stream2);
+  (void)hipStreamDestroy(ctx->stream3);
+  (void)hipEventDestroy(ctx->ev_blob);
+  (void)hipEventDestroy(ctx->ev_sub);
   delete ctx;
 }
 
@@ -820,8 +832,8 @@ static void tev_ensure(kv_ctx *ctx) {
   for (int i = 0; i < 12; i++) (void)hipEventCreate(&ctx->tev[i]);
   ctx->tev_init = true;
 }
-static inline void tev_rec(kv_ctx *ctx, int i) {
-  (void)hipEventRecord(ctx->tev[i], ctx->stream);
+static inline void tev_rec(kv_ctx *ctx, int i, hipStream_t stream = nullptr) {
+  (void)hipEventRecord(ctx->tev[i], stream ? stream : ctx->stream);
 }
 static double tev_ms(kv_ctx *ctx, int pair) {
   float ms = 0.f;
@@ -829,6 +841,92 @@ static double tev_ms(kv_ctx *ctx, int pair) {
       hipSuccess)
     return 0.0;
   return (double)ms;
+}
+
+/* Enqueue the muhash element + reduce chain for txs with include[t] != 0 on
+ * `stream`, fully async — the caller syncs the stream before reading outp.
+ * The blob must already be resident in g_vb.blob. With no work the identity
+ * partial is written synchronously and *launched stays false. */
+static int enqueue_muhash(kv_ctx *ctx, const std::vector<HTx> &txs,
+                          const uint8_t *include, uint64_t block_daa_score,
+                          hipStream_t stream, uint8_t *outp, bool *launched,
+                          std::vector<kv::kv_elem_job> &jobs) {
+  jobs.clear();
+  for (size_t t = 0; t < txs.size(); t++) {
+    if (!include[t]) continue;
+    const HTx &tx = txs[t];
+    uint8_t cb = h_is_coinbase(tx) ? 1 : 0;
+    for (uint32_t i = 0; i < tx.outputs.size(); i++)
+      jobs.push_back(kv::kv_elem_job{(uint32_t)t, tx.output_offs[i], i, 1, cb, 0,
+                                     block_daa_score});
+  }
+  size_t n_num = jobs.size();
+  for (size_t t = 0; t < txs.size(); t++) {
+    if (!include[t]) continue;
+    const HTx &tx = txs[t];
+    uint8_t cb = h_is_coinbase(tx) ? 1 : 0;
+    for (auto &in : tx.inputs)
+      jobs.push_back(kv::kv_elem_job{(uint32_t)t, in.rec_off, 0, 0, cb, 0,
+                                     block_daa_score});
+  }
+  size_t n_all = jobs.size(), n_den = n_all - n_num;
+  kv::u3072 one;
+  kv::u3072_one(one);
+  *launched = false;
+  if (n_all == 0) {
+    memcpy(outp, one.l, 384);
+    memcpy(outp + 384, one.l, 384);
+    return 0;
+  }
+  if (g_vb.elem_jobs.ensure(n_all * sizeof(kv::kv_elem_job)) ||
+      g_vb.elements.ensure(n_all * KVU_LIMBS * 8) ||
+      g_vb.partials_a.ensure(1024 * KVU_LIMBS * 8) ||
+      g_vb.partials_b.ensure(1024 * KVU_LIMBS * 8))
+    return -2;
+  HIP_CHECK(hipMemcpyAsync(g_vb.elem_jobs.p, jobs.data(),
+                           n_all * sizeof(kv::kv_elem_job), hipMemcpyHostToDevice,
+                           stream));
+  tev_rec(ctx, 10, stream);
+  hipLaunchKernelGGL(kv::kv_muhash_element_kernel,
+                     dim3(((uint32_t)n_all + 255) / 256), dim3(256), 0, stream,
+                     (const uint8_t *)g_vb.blob.p,
+                     (const kv::kv_elem_job *)g_vb.elem_jobs.p, (uint32_t)n_all,
+                     (uint64_t *)g_vb.elements.p);
+  /* reduce numerator then denominator halves down to one value each; the
+   * stride schedule is pure host arithmetic, so the whole chain enqueues
+   * without any device->host round trip */
+  for (int half = 0; half < 2; half++) {
+    size_t cnt = half == 0 ? n_num : n_den;
+    uint64_t *src = (uint64_t *)g_vb.elements.p + (half == 0 ? 0 : n_num * KVU_LIMBS);
+    uint64_t *pa = (uint64_t *)g_vb.partials_a.p;
+    uint64_t *pb = (uint64_t *)g_vb.partials_b.p;
+    if (cnt == 0) {
+      memcpy(outp + half * 384, one.l, 384);
+      continue;
+    }
+    uint32_t n_cur = (uint32_t)cnt;
+    uint64_t *cur = src;
+    while (n_cur > 1) {
+      /* wave-cooperative mulmod (one value per wave, limbs in registers).
+       * Keep every pass parallel: each wave chains <=64 values, so the final
+       * pass never degenerates into one wave grinding 512 serial mulmods. */
+      uint32_t stride = n_cur > 64 ? (n_cur + 63) / 64 : 1;
+      if (stride > 1024) stride = 1024;
+      uint32_t waves_per_block = 4; /* 256 threads */
+      uint32_t blocks = (stride + waves_per_block - 1) / waves_per_block;
+      hipLaunchKernelGGL(kv_u3072_reduce_wave_kernel, dim3(blocks), dim3(256), 0,
+                         stream, cur, n_cur, stride, pa);
+      n_cur = stride;
+      cur = pa;
+      std::swap(pa, pb);
+    }
+    HIP_CHECK(hipMemcpyAsync(outp + half * 384, cur, 384, hipMemcpyDeviceToHost,
+                             stream));
+  }
+  tev_rec(ctx, 11, stream);
+  HIP_CHECK(hipGetLastError());
+  *launched = true;
+  return 0;
 }
 
 /* caller holds ctx->mu; pre_codes (optional) pre-fails txs (e.g. missing
@@ -1051,12 +1149,13 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
    * misses only when the cache is on) */
   size_t ns = sjobs.size(), ne = ejobs.size();
   std::vector<uint8_t> s_gpu(ns), e_gpu(ne);
-  bool blob_uploaded = false;
+  bool blob_uploaded = false, subhash_done = false;
   if (ns + ne > 0) {
     if (g_vb.blob.ensure(blob_len) || g_vb.subhashes.ensure((size_t)n_txs * 160))
       return -2;
     HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
                              ctx->stream));
+    HIP_CHECK(hipEventRecord(ctx->ev_blob, ctx->stream));
     blob_uploaded = true;
     tev_rec(ctx, 0);
     hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256), dim3(256),
@@ -1064,6 +1163,8 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
                        (uint8_t *)g_vb.subhashes.p);
     tev_rec(ctx, 1);
     tev_rec_pair[0] = true;
+    HIP_CHECK(hipEventRecord(ctx->ev_sub, ctx->stream));
+    subhash_done = true;
     if (ns) {
       if (g_vb.s_jobs.ensure(ns * sizeof(kv::kv_job)) ||
           g_vb.s_tuples.ensure(ns * 128) || g_vb.s_bitmap.ensure((ns + 63) / 64 * 8) ||
@@ -1091,33 +1192,64 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
                                hipMemcpyDeviceToHost, ctx->stream));
     }
     if (ne) {
+      /* ecdsa chain on stream2: at block-path sizes both verify kernels are
+       * tiny latency-bound dispatches (a few hundred waves on a 2048-wave
+       * machine) — overlapping them hides the whole ecdsa chain behind the
+       * schnorr one */
       if (g_vb.e_jobs.ensure(ne * sizeof(kv::kv_job)) ||
           g_vb.e_tuples.ensure(ne * 132) || g_vb.e_bitmap.ensure((ne + 63) / 64 * 8) ||
           g_vb.e_status.ensure(ne))
         return -2;
+      HIP_CHECK(hipStreamWaitEvent(ctx->stream2, ctx->ev_sub, 0));
       HIP_CHECK(hipMemcpyAsync(g_vb.e_jobs.p, ejobs.data(), ne * sizeof(kv::kv_job),
-                               hipMemcpyHostToDevice, ctx->stream));
-      tev_rec(ctx, 4);
+                               hipMemcpyHostToDevice, ctx->stream2));
+      tev_rec(ctx, 4, ctx->stream2);
       hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
-                         dim3(((uint32_t)ne + 255) / 256), dim3(256), 0, ctx->stream,
+                         dim3(((uint32_t)ne + 255) / 256), dim3(256), 0, ctx->stream2,
                          (const uint8_t *)g_vb.blob.p, (const uint8_t *)g_vb.subhashes.p,
                          (const kv::kv_job *)g_vb.e_jobs.p, (uint32_t)ne,
                          (uint8_t *)g_vb.e_tuples.p, (uint8_t *)g_vb.e_tuples.p);
-      tev_rec(ctx, 5);
+      tev_rec(ctx, 5, ctx->stream2);
       tev_rec_pair[2] = true;
-      tev_rec(ctx, 8);
+      tev_rec(ctx, 8, ctx->stream2);
       hipLaunchKernelGGL(kv::kv_ecdsa_verify_kernel, dim3(((uint32_t)ne + 255) / 256),
-                         dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.e_tuples.p,
+                         dim3(256), 0, ctx->stream2, (const uint8_t *)g_vb.e_tuples.p,
                          (unsigned long long)ne, (unsigned long long *)g_vb.e_bitmap.p,
                          (uint8_t *)g_vb.e_status.p);
-      tev_rec(ctx, 9);
+      tev_rec(ctx, 9, ctx->stream2);
       tev_rec_pair[4] = true;
       ctx->last_timings.n_ecdsa += ne;
       HIP_CHECK(hipMemcpyAsync(e_gpu.data(), g_vb.e_status.p, ne,
-                               hipMemcpyDeviceToHost, ctx->stream));
+                               hipMemcpyDeviceToHost, ctx->stream2));
     }
     HIP_CHECK(hipGetLastError());
+  }
+
+  /* optimistic muhash on stream3, overlapped with the verify chains: include
+   * every tx that passed the phase-1 integer checks. Phase 4 consumes the
+   * result when the final accept set matches (the overwhelmingly common
+   * case) and re-enqueues the exact set otherwise. */
+  bool mu_opt_launched = false;
+  std::vector<uint8_t> mu_inc;
+  std::vector<kv::kv_elem_job> mu_jobs;
+  if (muhash_partial_out) {
+    mu_inc.resize(n_txs);
+    for (int t = 0; t < n_txs; t++) mu_inc[t] = codes[t] == 0;
+    if (!blob_uploaded) {
+      if (g_vb.blob.ensure(blob_len)) return -2;
+      HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
+                               ctx->stream));
+      HIP_CHECK(hipEventRecord(ctx->ev_blob, ctx->stream));
+      blob_uploaded = true;
+    }
+    HIP_CHECK(hipStreamWaitEvent(ctx->stream3, ctx->ev_blob, 0));
+    int mrc = enqueue_muhash(ctx, txs, mu_inc.data(), block_daa_score, ctx->stream3,
+                             muhash_partial_out, &mu_opt_launched, mu_jobs);
+    if (mrc) return mrc;
+  }
+  if (ns + ne > 0) {
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream2));
   }
 
   /* scatter GPU statuses back and remember fresh verdicts */
@@ -1201,16 +1333,18 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
       std::vector<uint8_t> s_st(ns_i), e_st(ne_i);
       if (ns_i + ne_i > 0) {
         if (!blob_uploaded) {
-          if (g_vb.blob.ensure(blob_len) ||
-              g_vb.subhashes.ensure((size_t)n_txs * 160))
-            return -2;
+          if (g_vb.blob.ensure(blob_len)) return -2;
           HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len,
                                    hipMemcpyHostToDevice, ctx->stream));
+          blob_uploaded = true;
+        }
+        if (!subhash_done) {
+          if (g_vb.subhashes.ensure((size_t)n_txs * 160)) return -2;
           hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256),
                              dim3(256), 0, ctx->stream,
                              (const uint8_t *)g_vb.blob.p, (uint32_t)n_txs,
                              (uint8_t *)g_vb.subhashes.p);
-          blob_uploaded = true;
+          subhash_done = true;
         }
         if (ns_i) {
           if (g_vb.s_tuples.ensure(st_h.size()) ||
@@ -1330,85 +1464,27 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     }
   });
 
-  /* phase 4: muhash over valid txs (spends → denominator, creates → numerator) */
+  /* phase 4: consume the optimistic muhash, or re-enqueue over the exact
+   * accept set when some tx failed validation after the integer checks */
   if (muhash_partial_out) {
-    std::vector<kv::kv_elem_job> njobs_v, djobs_v;
-    for (int t = 0; t < n_txs; t++) {
-      if (codes[t]) continue;
-      const HTx &tx = txs[t];
-      uint8_t cb = h_is_coinbase(tx) ? 1 : 0;
-      for (auto &in : tx.inputs)
-        djobs_v.push_back(kv::kv_elem_job{(uint32_t)t, in.rec_off, 0, 0, cb, 0,
-                                          block_daa_score});
-      for (uint32_t i = 0; i < tx.outputs.size(); i++)
-        njobs_v.push_back(kv::kv_elem_job{(uint32_t)t, tx.output_offs[i], i, 1, cb, 0,
-                                          block_daa_score});
-    }
-    size_t n_num = njobs_v.size(), n_den = djobs_v.size();
-    size_t n_all = n_num + n_den;
-    kv::u3072 one;
-    kv::u3072_one(one);
-    uint8_t *outp = muhash_partial_out;
-    if (n_all == 0) {
-      memcpy(outp, one.l, 384);
-      memcpy(outp + 384, one.l, 384);
-    } else {
-      std::vector<kv::kv_elem_job> all;
-      all.reserve(n_all);
-      all.insert(all.end(), njobs_v.begin(), njobs_v.end());
-      all.insert(all.end(), djobs_v.begin(), djobs_v.end());
-      if (g_vb.blob.ensure(blob_len) ||
-          g_vb.elem_jobs.ensure(n_all * sizeof(kv::kv_elem_job)) ||
-          g_vb.elements.ensure(n_all * KVU_LIMBS * 8) ||
-          g_vb.partials_a.ensure(1024 * KVU_LIMBS * 8) ||
-          g_vb.partials_b.ensure(1024 * KVU_LIMBS * 8))
-        return -2;
-      if (!blob_uploaded) /* phase 2 may have uploaded it already */
-        HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
-                                 ctx->stream));
-      HIP_CHECK(hipMemcpyAsync(g_vb.elem_jobs.p, all.data(),
-                               n_all * sizeof(kv::kv_elem_job), hipMemcpyHostToDevice,
-                               ctx->stream));
-      tev_rec(ctx, 10);
-      hipLaunchKernelGGL(kv::kv_muhash_element_kernel,
-                         dim3(((uint32_t)n_all + 255) / 256), dim3(256), 0, ctx->stream,
-                         (const uint8_t *)g_vb.blob.p,
-                         (const kv::kv_elem_job *)g_vb.elem_jobs.p, (uint32_t)n_all,
-                         (uint64_t *)g_vb.elements.p);
-      /* reduce numerator then denominator halves down to one value each */
-      for (int half = 0; half < 2; half++) {
-        size_t cnt = half == 0 ? n_num : n_den;
-        uint64_t *src = (uint64_t *)g_vb.elements.p + (half == 0 ? 0 : n_num * KVU_LIMBS);
-        uint64_t *pa = (uint64_t *)g_vb.partials_a.p;
-        uint64_t *pb = (uint64_t *)g_vb.partials_b.p;
-        if (cnt == 0) {
-          memcpy(outp + half * 384, one.l, 384);
-          continue;
-        }
-        uint32_t n_cur = (uint32_t)cnt;
-        uint64_t *cur = src;
-        while (n_cur > 1) {
-          /* wave-cooperative mulmod (one value per wave, limbs in registers).
-           * Keep every pass parallel: each wave chains ≤64 values, so the
-           * final pass never degenerates into one wave grinding 512 serial
-           * mulmods (measured 2.5 ms at the old stride schedule). */
-          uint32_t stride = n_cur > 64 ? (n_cur + 63) / 64 : 1;
-          if (stride > 1024) stride = 1024;
-          uint32_t waves_per_block = 4; /* 256 threads */
-          uint32_t blocks = (stride + waves_per_block - 1) / waves_per_block;
-          hipLaunchKernelGGL(kv_u3072_reduce_wave_kernel, dim3(blocks), dim3(256),
-                             0, ctx->stream, cur, n_cur, stride, pa);
-          n_cur = stride;
-          cur = pa;
-          std::swap(pa, pb);
-        }
-        HIP_CHECK(hipGetLastError());
-        HIP_CHECK(hipMemcpyAsync(outp + half * 384, cur, 384, hipMemcpyDeviceToHost,
-                                 ctx->stream));
-        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    bool match = true;
+    for (int t = 0; t < n_txs; t++)
+      if ((codes[t] == 0) != (mu_inc[t] != 0)) {
+        match = false;
+        break;
       }
-      tev_rec(ctx, 11);
-      tev_rec_pair[5] = true;
+    HIP_CHECK(hipStreamSynchronize(ctx->stream3)); /* optimistic writes done */
+    if (!match) {
+      for (int t = 0; t < n_txs; t++) mu_inc[t] = codes[t] == 0;
+      bool relaunched = false;
+      int mrc = enqueue_muhash(ctx, txs, mu_inc.data(), block_daa_score,
+                               ctx->stream, muhash_partial_out, &relaunched,
+                               mu_jobs);
+      if (mrc) return mrc;
+      HIP_CHECK(hipStreamSynchronize(ctx->stream));
+      tev_rec_pair[5] = relaunched;
+    } else {
+      tev_rec_pair[5] = mu_opt_launched;
     }
   }
 

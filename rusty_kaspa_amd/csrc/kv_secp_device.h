@@ -754,7 +754,7 @@ __device__ __forceinline__ void gej_cmov(gej &r, const gej &a, u64 cond) {
 }
 
 /* mixed add (b affine, magnitude <= 3 after phi/negation); a coords mag <= 2 */
-__device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
+__device__ __forceinline__ void gej_add_ge_impl(gej &r, const gej &a, const ge &b) {
   u64 a_inf = (u64)fe26_is_zero(a.z);
   fe26 z1z1, u2, s2, h, hh, i, j, rr, v, t;
   fe26_sqr(z1z1, a.z);       /* 1 */
@@ -830,6 +830,10 @@ __device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
 }
 
 /* full add; a and b coords magnitude <= 3 */
+__device__ KV_GROUP_ATTR void gej_add_ge(gej &r, const gej &a, const ge &b) {
+  gej_add_ge_impl(r, a, b);
+}
+
 __device__ KV_GROUP_ATTR void gej_add(gej &r, const gej &a, const gej &b) {
   u64 a_inf = (u64)fe26_is_zero(a.z);
   fe26 z1z1, z2z2, u1, u2, s1, s2, h, i, j, rr, v, t;

@@ -246,6 +246,66 @@ __device__ __forceinline__ u64 glv_digit(const glv_half &h, int w) {
   return v & 15;
 }
 
+/* One full ladder window in ONE call frame: 4 doublings + the 4 stream adds
+ * (G, φG, P, φP). The group ops inline INSIDE this body, so the accumulator
+ * crosses the noinline ABI once per window instead of five times — the r02
+ * PMC showed the per-call scratch spills of R/temps were the kernel's
+ * dominant memory traffic (~119KB/verify). Full kernel-wide inlining is not
+ * an option: it reproducibly hangs gfx950 (measured again this round); this
+ * bounded body (~5k instructions) stays under that cliff. */
+__device__ KV_GROUP_ATTR void gej_window_step(gej &R, const ge *ptab,
+                                              const fe26 &beta, u64 dg1, u64 ng1,
+                                              u64 dg2, u64 ng2, u64 dp1, u64 np1,
+                                              u64 dp2, u64 np2) {
+  gej t;
+  gej_double_impl(t, R);
+  gej_double_impl(R, t);
+  gej_double_impl(t, R);
+  gej_double_impl(R, t);
+  /* G stream */
+  {
+    ge e = KV_G_TABLE[dg1];
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)ng1);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(dg1 != 0));
+  }
+  /* φG stream: (β·x, ±y) */
+  {
+    ge e = KV_G_TABLE[dg2];
+    fe26 bx;
+    fe26_mul(bx, e.x, beta);
+    e.x = bx;
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)ng2);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(dg2 != 0));
+  }
+  /* P stream (mixed add vs the affine per-lane table) */
+  {
+    ge e = ptab[dp1];
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)np1);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(dp1 != 0));
+  }
+  /* φP stream: (β·x, ±y) */
+  {
+    ge e = ptab[dp2];
+    fe26 bx;
+    fe26_mul(bx, e.x, beta);
+    e.x = bx;
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)np2);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(dp2 != 0));
+  }
+}
+
 /* R = gs·G + ps·P via GLV-split 4-bit windows: 33 window steps of 4 doublings
  * + 4 selected adds (G, φG, P, φP streams; φ applied at add time as one β·x
  * field multiply; negative half-scalars negate the added point's y). */
@@ -313,54 +373,11 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
   gej_set_infinity(R);
 #pragma unroll 1
   for (int w = 32; w >= 0; w--) {
-    gej t;
-    gej_double4(R, R);
-    /* G stream */
-    {
-      u64 d = glv_digit(g1h, w);
-      ge e = KV_G_TABLE[d];
-      fe26 ny;
-      fe26_neg(ny, e.y, 2);
-      fe26_cmov(e.y, ny, (u32)g1h.neg);
-      gej_add_ge(t, R, e);
-      gej_cmov(R, t, (u64)(d != 0));
-    }
-    /* φG stream: (β·x, ±y) */
-    {
-      u64 d = glv_digit(g2h, w);
-      ge e = KV_G_TABLE[d];
-      fe26 bx;
-      fe26_mul(bx, e.x, beta);
-      e.x = bx;
-      fe26 ny;
-      fe26_neg(ny, e.y, 2);
-      fe26_cmov(e.y, ny, (u32)g2h.neg);
-      gej_add_ge(t, R, e);
-      gej_cmov(R, t, (u64)(d != 0));
-    }
-    /* P stream (mixed add vs the affine table) */
-    {
-      u64 d = glv_digit(p1h, w);
-      ge e = ptab[d];
-      fe26 ny;
-      fe26_neg(ny, e.y, 2);
-      fe26_cmov(e.y, ny, (u32)p1h.neg);
-      gej_add_ge(t, R, e);
-      gej_cmov(R, t, (u64)(d != 0));
-    }
-    /* φP stream: (β·x, ±y) */
-    {
-      u64 d = glv_digit(p2h, w);
-      ge e = ptab[d];
-      fe26 bx;
-      fe26_mul(bx, e.x, beta);
-      e.x = bx;
-      fe26 ny;
-      fe26_neg(ny, e.y, 2);
-      fe26_cmov(e.y, ny, (u32)p2h.neg);
-      gej_add_ge(t, R, e);
-      gej_cmov(R, t, (u64)(d != 0));
-    }
+    gej_window_step(R, ptab, beta,
+                    glv_digit(g1h, w), g1h.neg,
+                    glv_digit(g2h, w), g2h.neg,
+                    glv_digit(p1h, w), p1h.neg,
+                    glv_digit(p2h, w), p2h.neg);
   }
 }
 
