@@ -217,16 +217,41 @@ int kv_fetch_bitmap(kv_ctx *ctx, size_t n, uint64_t *bitmap_out);
  * ⇔ UtxoCollection (consensus/core/src/utxo/utxo_collection.rs:5) + the
  * populate/diff steps (utxo_validation.rs:351-390, utxo_diff.rs:224).
  * Outpoints are 36B (tx_id‖index LE); entries are packed 64B records:
- * amount u64 ‖ daa_score u64 ‖ flags u16 (bit0 coinbase) ‖ spk_version u16 ‖
- * spk_len u32 ‖ spk[36] inline (standard SPKs ≤ 35B; larger scripts are a
- * documented round-2 arena extension). */
+ * amount u64 ‖ daa_score u64 ‖ flags u16 ‖ spk_version u16 ‖
+ * spk_len u32 ‖ spk[36] inline when spk_len ≤ 36.
+ * Scripts larger than 36B (the reference allows spk up to
+ * max_script_public_key_len = 10,000B, utxo_entry.rs:20 / params.rs) live in
+ * a device-side ARENA: the entry carries flags bit1
+ * (KV_UTXO_F_SPK_ARENA) and its spk field's first 4 bytes are the u32 arena
+ * byte offset. Arena space is bump-allocated; removing an entry leaks its
+ * arena span until the next kv_utxo_reset (documented trade: non-standard
+ * outputs are rare and the arena compacts on reset). */
+#define KV_UTXO_F_COINBASE 1u
+#define KV_UTXO_F_SPK_ARENA 2u
+#define KV_UTXO_MAX_SPK 10000u
 int kv_utxo_reset(kv_ctx *ctx, uint64_t capacity);
+/* inline-only upsert: every entry's spk_len must be ≤ 36 */
 int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints, const uint8_t *entries64,
                    size_t n);
+/* general upsert: entries with spk_len > 36 take their script bytes from
+ * spk_blob, concatenated in entry order (total = sum of those spk_lens);
+ * the engine stores them in the arena and rewrites the entry in the table
+ * with KV_UTXO_F_SPK_ARENA + the arena offset. */
+int kv_utxo_upsert_spk(kv_ctx *ctx, const uint8_t *outpoints,
+                       const uint8_t *entries64, const uint8_t *spk_blob,
+                       size_t spk_blob_len, size_t n);
 int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n);
 int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
                    uint8_t *entries_out /*[64n] or NULL*/, uint64_t *found_bitmap,
                    double *kernel_ms /*optional*/);
+/* lookup that also returns out-of-line scripts: arena entries' bytes are
+ * gathered (device-side) into spk_out in entry order and each such returned
+ * entry's spk field is rewritten to the u32 offset INTO spk_out. *spk_used
+ * receives the total bytes (call with spk_cap 0 to size). Returns -3 when
+ * spk_cap is too small (entries_out/bitmap still valid, *spk_used = needed). */
+int kv_utxo_lookup_spk(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
+                       uint8_t *entries_out, uint64_t *found_bitmap,
+                       uint8_t *spk_out, size_t spk_cap, size_t *spk_used);
 
 /* Populate + validate + (optionally) apply the UTXO diff — the full virtual
  * processor step for one block (utxo_validation.rs:351-390 populate, then

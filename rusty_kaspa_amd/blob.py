@@ -146,9 +146,12 @@ def strip_utxo_entries(blob: bytes):
             spkv, = struct.unpack_from("<H", blob, p + 18)
             spk_len, = struct.unpack_from("<I", blob, p + 20)
             spk = blob[p + 24:p + 24 + spk_len]
-            assert spk_len <= 36, "inline-table spk limit"
+            # seeds carry inline spks only; long scripts (spk_len > 36) go
+            # through kv_utxo_upsert_spk — their seed entry keeps the real
+            # spk_len with a zeroed inline area and the caller supplies the
+            # bytes via the spk blob (tests do this explicitly)
             entry64 = struct.pack("<QQHHI", amount, daa, is_cb & 1, spkv,
-                                  spk_len) + spk.ljust(36, b"\0") + bytes(4)
+                                  spk_len) + spk[:36].ljust(36, b"\0") + bytes(4)
             seeds.append((outpoint, entry64))
             chunks.append(bytes(24))  # zero entry, spk_len 0
             p += 24 + spk_len + (32 if has_cov else 0)

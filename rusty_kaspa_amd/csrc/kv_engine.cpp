@@ -91,6 +91,13 @@ struct kv_ctx {
   /* GPU-resident UTXO set */
   kv::utxo_slot *d_utxo = nullptr;
   uint64_t utxo_cap = 0; /* power of two */
+  /* out-of-line spk arena (entries with spk_len > 36; flags bit1) */
+  uint8_t *d_arena = nullptr;
+  uint64_t arena_cap = 0, arena_head = 0;
+  void *d_gjobs = nullptr;
+  size_t d_gjobs_cap = 0;
+  void *d_gout = nullptr;
+  size_t d_gout_cap = 0;
   uint8_t *d_op_in = nullptr;
   size_t d_op_cap = 0;
   uint8_t *d_val_in = nullptr;
@@ -152,6 +159,9 @@ extern "C" void kv_destroy(kv_ctx *ctx) {
   if (ctx->d_in) (void)hipFree(ctx->d_in);
   if (ctx->d_bitmap) (void)hipFree(ctx->d_bitmap);
   if (ctx->d_status) (void)hipFree(ctx->d_status);
+  if (ctx->d_arena) (void)hipFree(ctx->d_arena);
+  if (ctx->d_gjobs) (void)hipFree(ctx->d_gjobs);
+  if (ctx->d_gout) (void)hipFree(ctx->d_gout);
   (void)hipStreamDestroy(ctx->stream);
   (void)hipStreamDestroy(ctx->stream2);
   (void)hipStreamDestroy(ctx->stream3);
@@ -1539,6 +1549,7 @@ extern "C" int kv_utxo_reset(kv_ctx *ctx, uint64_t capacity) {
   HIP_CHECK(hipMemsetAsync(ctx->d_utxo, 0, cap * sizeof(kv::utxo_slot), ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
   ctx->utxo_cap = cap;
+  ctx->arena_head = 0; /* the spk arena compacts on reset (allocation kept) */
   return 0;
 }
 
@@ -1585,6 +1596,88 @@ extern "C" int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints,
   return utxo_upsert_nolock(ctx, outpoints, entries64, n);
 }
 
+/* grow the spk arena to fit `need` more bytes; preserves contents */
+static int arena_reserve(kv_ctx *ctx, uint64_t need) {
+  uint64_t want = ctx->arena_head + need;
+  if (want <= ctx->arena_cap) return 0;
+  uint64_t nc = ctx->arena_cap ? ctx->arena_cap : (1u << 20);
+  while (nc < want) nc *= 2;
+  uint8_t *na = nullptr;
+  if (hipMalloc(&na, nc) != hipSuccess) {
+    set_error("kv_utxo arena: hipMalloc failed");
+    return -2;
+  }
+  if (ctx->d_arena && ctx->arena_head)
+    HIP_CHECK(hipMemcpyAsync(na, ctx->d_arena, ctx->arena_head,
+                             hipMemcpyDeviceToDevice, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  if (ctx->d_arena) (void)hipFree(ctx->d_arena);
+  ctx->d_arena = na;
+  ctx->arena_cap = nc;
+  return 0;
+}
+
+/* general upsert: spk_len > 36 entries spill their scripts to the arena
+ * (bump-allocated; spans leak on remove until kv_utxo_reset — documented) */
+static int utxo_upsert_spk_nolock(kv_ctx *ctx, const uint8_t *outpoints,
+                                  const uint8_t *entries64, const uint8_t *spk_blob,
+                                  size_t spk_blob_len, size_t n) {
+  if (!ctx->d_utxo) {
+    set_error("kv_utxo_upsert_spk: call kv_utxo_reset first");
+    return -1;
+  }
+  uint64_t long_total = 0;
+  for (size_t i = 0; i < n; i++) {
+    uint32_t spk_len;
+    memcpy(&spk_len, entries64 + i * 64 + 20, 4);
+    if (spk_len > 36) {
+      if (spk_len > KV_UTXO_MAX_SPK) {
+        set_error("kv_utxo_upsert_spk: spk exceeds KV_UTXO_MAX_SPK");
+        return -1;
+      }
+      long_total += spk_len;
+    }
+  }
+  if (long_total > spk_blob_len) {
+    set_error("kv_utxo_upsert_spk: spk_blob shorter than the long entries");
+    return -1;
+  }
+  if (long_total == 0)
+    return utxo_upsert_nolock(ctx, outpoints, entries64, n);
+  int rc = arena_reserve(ctx, long_total);
+  if (rc) return rc;
+  uint64_t base = ctx->arena_head;
+  HIP_CHECK(hipMemcpyAsync(ctx->d_arena + base, spk_blob, long_total,
+                           hipMemcpyHostToDevice, ctx->stream));
+  ctx->arena_head += long_total;
+  /* rewrite the long entries: flags |= ARENA, spk[0..4) = arena offset */
+  std::vector<uint8_t> ents(entries64, entries64 + n * 64);
+  uint64_t off = base;
+  for (size_t i = 0; i < n; i++) {
+    uint8_t *e = ents.data() + i * 64;
+    uint32_t spk_len;
+    memcpy(&spk_len, e + 20, 4);
+    if (spk_len <= 36) continue;
+    uint16_t flags;
+    memcpy(&flags, e + 16, 2);
+    flags |= KV_UTXO_F_SPK_ARENA;
+    memcpy(e + 16, &flags, 2);
+    uint32_t o32 = (uint32_t)off;
+    memcpy(e + 24, &o32, 4);
+    memset(e + 28, 0, 32);
+    off += spk_len;
+  }
+  return utxo_upsert_nolock(ctx, outpoints, ents.data(), n);
+}
+
+extern "C" int kv_utxo_upsert_spk(kv_ctx *ctx, const uint8_t *outpoints,
+                                  const uint8_t *entries64, const uint8_t *spk_blob,
+                                  size_t spk_blob_len, size_t n) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  return utxo_upsert_spk_nolock(ctx, outpoints, entries64, spk_blob,
+                                spk_blob_len, n);
+}
+
 static int utxo_remove_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
   if (!ctx->d_utxo) {
     set_error("kv_utxo_remove: call kv_utxo_reset first");
@@ -1603,6 +1696,78 @@ static int utxo_remove_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
 extern "C" int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
   std::lock_guard<std::mutex> lk(ctx->mu);
   return utxo_remove_nolock(ctx, outpoints, n);
+}
+
+static int utxo_lookup_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
+                              uint8_t *entries_out, uint64_t *found_bitmap,
+                              double *kernel_ms);
+
+/* gather the arena spans of entries flagged KV_UTXO_F_SPK_ARENA (in entry
+ * order) into host memory, rewriting each entry's spk field to its offset in
+ * the gathered buffer. `take(i)` -> entry pointer or NULL to skip. */
+template <class TakeFn>
+static int arena_gather_nolock(kv_ctx *ctx, size_t n, TakeFn take,
+                               std::vector<uint8_t> &out) {
+  std::vector<kv::arena_gather_job> jobs;
+  uint32_t dst = 0;
+  for (size_t i = 0; i < n; i++) {
+    uint8_t *e = take(i);
+    if (!e) continue;
+    uint16_t flags;
+    memcpy(&flags, e + 16, 2);
+    if (!(flags & KV_UTXO_F_SPK_ARENA)) continue;
+    uint32_t spk_len, src;
+    memcpy(&spk_len, e + 20, 4);
+    memcpy(&src, e + 24, 4);
+    jobs.push_back(kv::arena_gather_job{src, spk_len, dst, 0});
+    memcpy(e + 24, &dst, 4); /* rewrite to the gathered-buffer offset */
+    dst += spk_len;
+  }
+  out.resize(dst);
+  if (jobs.empty()) return 0;
+  if (ensure_cap(&ctx->d_gjobs, &ctx->d_gjobs_cap,
+                 jobs.size() * sizeof(kv::arena_gather_job)) ||
+      ensure_cap(&ctx->d_gout, &ctx->d_gout_cap, dst))
+    return -2;
+  HIP_CHECK(hipMemcpyAsync(ctx->d_gjobs, jobs.data(),
+                           jobs.size() * sizeof(kv::arena_gather_job),
+                           hipMemcpyHostToDevice, ctx->stream));
+  hipLaunchKernelGGL(kv::kv_arena_gather_kernel, dim3((uint32_t)jobs.size()),
+                     dim3(256), 0, ctx->stream, ctx->d_arena,
+                     (const kv::arena_gather_job *)ctx->d_gjobs,
+                     (uint32_t)jobs.size(), (uint8_t *)ctx->d_gout);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipMemcpyAsync(out.data(), ctx->d_gout, dst, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return 0;
+}
+
+extern "C" int kv_utxo_lookup_spk(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
+                                  uint8_t *entries_out, uint64_t *found_bitmap,
+                                  uint8_t *spk_out, size_t spk_cap,
+                                  size_t *spk_used) {
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  int rc = utxo_lookup_nolock(ctx, outpoints, n, entries_out, found_bitmap,
+                              nullptr);
+  if (rc) return rc;
+  std::vector<uint8_t> gathered;
+  rc = arena_gather_nolock(
+      ctx, n,
+      [&](size_t i) -> uint8_t * {
+        int hit = (found_bitmap[i / 64] >> (i % 64)) & 1;
+        return hit ? entries_out + i * 64 : nullptr;
+      },
+      gathered);
+  if (rc) return rc;
+  if (spk_used) *spk_used = gathered.size();
+  if (gathered.empty()) return 0;
+  if (gathered.size() > spk_cap) {
+    set_error("kv_utxo_lookup_spk: spk_out too small");
+    return -3;
+  }
+  memcpy(spk_out, gathered.data(), gathered.size());
+  return 0;
 }
 
 static int utxo_lookup_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
@@ -1911,9 +2076,18 @@ static int populate_blob_nolock(kv_ctx *ctx, const uint8_t *blob, size_t blob_le
     return std::chrono::duration<double, std::milli>(b - a).count();
   };
   auto t_a = tnow();
+  std::vector<uint8_t> gathered; /* out-of-line spk bytes, entry order */
   if (n_in_total) {
     int rc = utxo_lookup_nolock(ctx, ops.data(), n_in_total, entries.data(),
                                 found.data(), nullptr);
+    if (rc) return rc;
+    rc = arena_gather_nolock(
+        ctx, n_in_total,
+        [&](size_t i) -> uint8_t * {
+          int hit = (found[i / 64] >> (i % 64)) & 1;
+          return hit ? entries.data() + i * 64 : nullptr;
+        },
+        gathered);
     if (rc) return rc;
   }
   auto t_b = tnow();
@@ -1986,7 +2160,13 @@ static int populate_blob_nolock(kv_ctx *ctx, const uint8_t *blob, size_t blob_le
       memcpy(dst + 20, &spk_len, 4);
       dst += 24;
       if (spk_len) {
-        memcpy(dst, e + 24, spk_len);
+        if (eflags & KV_UTXO_F_SPK_ARENA) {
+          uint32_t go;
+          memcpy(&go, e + 24, 4);
+          memcpy(dst, gathered.data() + go, spk_len);
+        } else {
+          memcpy(dst, e + 24, spk_len);
+        }
         dst += spk_len;
       }
       ii++;
@@ -2032,7 +2212,7 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
   if (rc || !apply_diff) return rc;
 
   /* diff apply for accepted txs: remove spent, upsert created */
-  std::vector<uint8_t> del_ops, add_ops, add_ents;
+  std::vector<uint8_t> del_ops, add_ops, add_ents, add_spk;
   for (int t = 0; t < n_txs; t++) {
     const HTx &tx = txs[t];
     if (tx_codes_out[t] != 0) continue;
@@ -2044,9 +2224,8 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
     }
     for (uint32_t i = 0; i < tx.outputs.size(); i++) {
       const HOutput &o = tx.outputs[i];
-      if (o.spk_len > 36) {
-        set_error("kv_validate_block_utxo: created spk > 36B (round-1 inline "
-                  "table limit)");
+      if (o.spk_len > KV_UTXO_MAX_SPK) {
+        set_error("kv_validate_block_utxo: created spk exceeds KV_UTXO_MAX_SPK");
         return -4;
       }
       add_ops.insert(add_ops.end(), tx.tx_id, tx.tx_id + 32);
@@ -2058,7 +2237,10 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
       /* flags: coinbase txs never enter this path → bit0 = 0 */
       memcpy(e + 18, &o.spk_version, 2);
       memcpy(e + 20, &o.spk_len, 4);
-      memcpy(e + 24, o.spk, o.spk_len);
+      if (o.spk_len <= 36)
+        memcpy(e + 24, o.spk, o.spk_len);
+      else /* out-of-line: bytes go to the arena via upsert_spk below */
+        add_spk.insert(add_spk.end(), o.spk, o.spk + o.spk_len);
       add_ents.insert(add_ents.end(), e, e + 64);
     }
   }
@@ -2067,8 +2249,9 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
     if (rc) return rc;
   }
   if (!add_ops.empty()) {
-    rc = utxo_upsert_nolock(ctx, add_ops.data(), add_ents.data(),
-                            add_ops.size() / 36);
+    rc = utxo_upsert_spk_nolock(ctx, add_ops.data(), add_ents.data(),
+                                add_spk.data(), add_spk.size(),
+                                add_ops.size() / 36);
     if (rc) return rc;
   }
   return 0;

@@ -174,4 +174,24 @@ extern "C" __global__ void kv_utxo_lookup_kernel(const utxo_slot *__restrict__ t
   if ((threadIdx.x & 63) == 0 && i < n) found[i / 64] = mask;
 }
 
+/* gather out-of-line scripts from the arena: one job per BLOCK, 256 threads
+ * striding the bytes (spans reach KV_UTXO_MAX_SPK = 10KB) */
+struct arena_gather_job {
+  uint32_t src_off; /* byte offset in the arena */
+  uint32_t len;
+  uint32_t dst_off; /* byte offset in out */
+  uint32_t _pad;
+};
+
+extern "C" __global__ void kv_arena_gather_kernel(const uint8_t *__restrict__ arena,
+                                                  const arena_gather_job *__restrict__ jobs,
+                                                  uint32_t n_jobs,
+                                                  uint8_t *__restrict__ out) {
+  uint32_t j = blockIdx.x;
+  if (j >= n_jobs) return;
+  arena_gather_job job = jobs[j];
+  for (uint32_t i = threadIdx.x; i < job.len; i += blockDim.x)
+    out[job.dst_off + i] = arena[job.src_off + i];
+}
+
 } // namespace kv
