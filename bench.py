@@ -63,11 +63,12 @@ ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 611,150
 # gfx950 VALU issue peak: 256 CU x 4 SIMD x 32 lanes x 2.4 GHz = 78.6 T u32/s
 VALU_PEAK_TOPS = 78.6
 # Memory-side traffic per verify, measured by rocprofv3 --pmc FETCH_SIZE /
-# WRITE_SIZE (separate passes) on 262144-tuple staged dispatches:
-# (1.352e7 + 1.702e7) KiB / 262144 = 119,296 B/verify (fetch 52.8KB + write
-# 66.5KB; dominated by the __noinline__ group-op scratch ABI, vs 128B of
-# algorithmic input). Provenance: profiles/r02_rocprof_summary.json.
-TRAFFIC_BYTES_PER_VERIFY = 119296
+# WRITE_SIZE (separate passes) on 262144-tuple staged dispatches of the
+# window-fused kernel: (8.417e6 + 5.626e6) KiB / 262144 = 54,856 B/verify
+# (fetch 32.9KB + write 22.0KB — scratch table reads + the once-per-window
+# accumulator spill; vs 128B of algorithmic input). Provenance:
+# profiles/r02b_verify_window_pmc.json (pre-fusion: 119,296).
+TRAFFIC_BYTES_PER_VERIFY = 54856
 
 
 def log(msg):

@@ -953,6 +953,13 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     return -1;
   }
   uint64_t sigop_units = ctx->params.mass_per_sig_op * KVH_UNITS_PER_GRAM;
+  const bool kv_timing = getenv("KV_TIMING") != nullptr;
+  auto vt_now = []() { return std::chrono::steady_clock::now(); };
+  auto vt_ms = [](std::chrono::steady_clock::time_point a,
+                  std::chrono::steady_clock::time_point b) {
+    return std::chrono::duration<double, std::milli>(b - a).count();
+  };
+  auto vt0 = vt_now();
   tev_ensure(ctx);
   bool tev_rec_pair[6] = {false, false, false, false, false, false};
   ctx->last_timings = kv_validate_timings{};
@@ -1071,6 +1078,8 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
         }
       }
   }
+
+  auto vt1 = vt_now();
 
   /* phase 1.5: sig cache ⇔ TransactionValidator sig_cache (caches.rs:57-82).
    * Key = blake2b-256(tx_id ‖ digest(all input entries) ‖ input_index ‖
@@ -1257,10 +1266,12 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
                              muhash_partial_out, &mu_opt_launched, mu_jobs);
     if (mrc) return mrc;
   }
+  auto vt2 = vt_now();
   if (ns + ne > 0) {
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
     HIP_CHECK(hipStreamSynchronize(ctx->stream2));
   }
+  auto vt3 = vt_now();
 
   /* scatter GPU statuses back and remember fresh verdicts */
   if (use_cache) {
@@ -1460,6 +1471,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     }
   }
 
+  auto vt4 = vt_now();
   /* phase 3: resolution (first failing input wins, sequential semantics);
    * read-only over the GPU statuses → fans over the host cores */
   kvh_parallel_for((uint32_t)n_txs, [&](uint32_t t) {
@@ -1474,6 +1486,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     }
   });
 
+  auto vt5 = vt_now();
   /* phase 4: consume the optimistic muhash, or re-enqueue over the exact
    * accept set when some tx failed validation after the integer checks */
   if (muhash_partial_out) {
@@ -1511,6 +1524,12 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     if (codes[t]) fees[t] = 0; /* fee defined only for accepted txs */
   memcpy(tx_codes_out, codes.data(), (size_t)n_txs * 4);
   memcpy(fees_out, fees.data(), (size_t)n_txs * 8);
+  if (kv_timing)
+    fprintf(stderr,
+            "[kv_timing] validate: parse+p1 %.2f cache+enq %.2f gpu-wait %.2f "
+            "interp %.2f resolve %.2f muhash-wait %.2f total %.2f ms\n",
+            vt_ms(vt0, vt1), vt_ms(vt1, vt2), vt_ms(vt2, vt3), vt_ms(vt3, vt4),
+            vt_ms(vt4, vt5), vt_ms(vt5, vt_now()), vt_ms(vt0, vt_now()));
   return 0;
 }
 
