@@ -93,6 +93,12 @@ struct kv_ctx {
    * the enqueue phase) */
   uint8_t *h_s_status = nullptr, *h_e_status = nullptr;
   size_t h_s_cap = 0, h_e_cap = 0;
+  /* per-call PINNED upload staging arena: hipMemcpyAsync from pageable host
+   * memory blocks the enqueue thread (synchronous staging inside HIP); all
+   * per-call H2D uploads bounce through this bump arena instead. Regions
+   * stay valid until the call's final stream syncs. */
+  uint8_t *h_stage = nullptr;
+  size_t h_stage_cap = 0, h_stage_off = 0;
   std::vector<uint64_t> found_buf;
   /* GPU-resident UTXO set */
   kv::utxo_slot *d_utxo = nullptr;
@@ -167,6 +173,7 @@ extern "C" void kv_destroy(kv_ctx *ctx) {
   if (ctx->d_status) (void)hipFree(ctx->d_status);
   if (ctx->h_s_status) (void)hipHostFree(ctx->h_s_status);
   if (ctx->h_e_status) (void)hipHostFree(ctx->h_e_status);
+  if (ctx->h_stage) (void)hipHostFree(ctx->h_stage);
   if (ctx->d_arena) (void)hipFree(ctx->d_arena);
   if (ctx->d_gjobs) (void)hipFree(ctx->d_gjobs);
   if (ctx->d_gout) (void)hipFree(ctx->d_gout);
@@ -861,6 +868,42 @@ static double tev_ms(kv_ctx *ctx, int pair) {
   return (double)ms;
 }
 
+/* reserve the per-call pinned staging arena (call while all streams are
+ * idle — start of a validate call); returns -2 on allocation failure */
+static int stage_reserve(kv_ctx *ctx, size_t need) {
+  ctx->h_stage_off = 0;
+  if (ctx->h_stage_cap >= need) return 0;
+  if (ctx->h_stage) (void)hipHostFree(ctx->h_stage);
+  size_t nc = need + need / 2 + 4096;
+  if (hipHostMalloc(&ctx->h_stage, nc) != hipSuccess) {
+    ctx->h_stage = nullptr;
+    ctx->h_stage_cap = 0;
+    set_error("hipHostMalloc failed (staging)");
+    return -2;
+  }
+  ctx->h_stage_cap = nc;
+  return 0;
+}
+
+/* copy src into the pinned arena and return the pinned pointer (NULL when the
+ * arena is exhausted — callers then fall back to the pageable source) */
+static const uint8_t *stage_push(kv_ctx *ctx, const void *src, size_t len) {
+  if (!ctx->h_stage || ctx->h_stage_off + len > ctx->h_stage_cap) return nullptr;
+  uint8_t *dst = ctx->h_stage + ctx->h_stage_off;
+  memcpy(dst, src, len);
+  ctx->h_stage_off += len;
+  return dst;
+}
+
+/* stage-then-upload: pinned when the arena has room, pageable otherwise */
+static inline int h2d_staged(kv_ctx *ctx, void *dst, const void *src, size_t len,
+                             hipStream_t stream) {
+  const uint8_t *p = stage_push(ctx, src, len);
+  HIP_CHECK(hipMemcpyAsync(dst, p ? (const void *)p : src, len,
+                           hipMemcpyHostToDevice, stream));
+  return 0;
+}
+
 /* Enqueue the muhash element + reduce chain for txs with include[t] != 0 on
  * `stream`, fully async — the caller syncs the stream before reading outp.
  * The blob must already be resident in g_vb.blob. With no work the identity
@@ -901,9 +944,9 @@ static int enqueue_muhash(kv_ctx *ctx, const std::vector<HTx> &txs,
       g_vb.partials_a.ensure(1024 * KVU_LIMBS * 8) ||
       g_vb.partials_b.ensure(1024 * KVU_LIMBS * 8))
     return -2;
-  HIP_CHECK(hipMemcpyAsync(g_vb.elem_jobs.p, jobs.data(),
-                           n_all * sizeof(kv::kv_elem_job), hipMemcpyHostToDevice,
-                           stream));
+  if (h2d_staged(ctx, g_vb.elem_jobs.p, jobs.data(),
+                 n_all * sizeof(kv::kv_elem_job), stream))
+    return -2;
   tev_rec(ctx, 10, stream);
   hipLaunchKernelGGL(kv::kv_muhash_element_kernel,
                      dim3(((uint32_t)n_all + 255) / 256), dim3(256), 0, stream,
@@ -971,6 +1014,15 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
   tev_ensure(ctx);
   bool tev_rec_pair[6] = {false, false, false, false, false, false};
   ctx->last_timings = kv_validate_timings{};
+  {
+    /* generous upper bound: blob + verify jobs + muhash jobs */
+    size_t n_units = 0;
+    for (auto &tx : txs) n_units += tx.inputs.size() + tx.outputs.size();
+    if (stage_reserve(ctx, blob_len + n_units * (sizeof(kv::kv_job) +
+                                                 sizeof(kv::kv_elem_job)) +
+                               (size_t)n_txs * 64 + 65536))
+      return -2;
+  }
 
   /* phase 1: host integer checks + classification, fanned over the host cores
    * (⇔ the reference's rayon pool) with per-chunk job lists so the GPU job
@@ -1180,8 +1232,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
   if (ns + ne > 0) {
     if (g_vb.blob.ensure(blob_len) || g_vb.subhashes.ensure((size_t)n_txs * 160))
       return -2;
-    HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
-                             ctx->stream));
+    if (h2d_staged(ctx, g_vb.blob.p, blob, blob_len, ctx->stream)) return -2;
     HIP_CHECK(hipEventRecord(ctx->ev_blob, ctx->stream));
     blob_uploaded = true;
     tev_rec(ctx, 0);
@@ -1197,8 +1248,9 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
           g_vb.s_tuples.ensure(ns * 128) || g_vb.s_bitmap.ensure((ns + 63) / 64 * 8) ||
           g_vb.s_status.ensure(ns))
         return -2;
-      HIP_CHECK(hipMemcpyAsync(g_vb.s_jobs.p, sjobs.data(), ns * sizeof(kv::kv_job),
-                               hipMemcpyHostToDevice, ctx->stream));
+      if (h2d_staged(ctx, g_vb.s_jobs.p, sjobs.data(), ns * sizeof(kv::kv_job),
+                     ctx->stream))
+        return -2;
       tev_rec(ctx, 2);
       hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                          dim3(((uint32_t)ns + 255) / 256), dim3(256), 0, ctx->stream,
@@ -1226,8 +1278,9 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
           g_vb.e_status.ensure(ne))
         return -2;
       HIP_CHECK(hipStreamWaitEvent(ctx->stream2, ctx->ev_sub, 0));
-      HIP_CHECK(hipMemcpyAsync(g_vb.e_jobs.p, ejobs.data(), ne * sizeof(kv::kv_job),
-                               hipMemcpyHostToDevice, ctx->stream2));
+      if (h2d_staged(ctx, g_vb.e_jobs.p, ejobs.data(), ne * sizeof(kv::kv_job),
+                     ctx->stream2))
+        return -2;
       tev_rec(ctx, 4, ctx->stream2);
       hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                          dim3(((uint32_t)ne + 255) / 256), dim3(256), 0, ctx->stream2,
@@ -1260,8 +1313,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     for (int t = 0; t < n_txs; t++) mu_inc[t] = codes[t] == 0;
     if (!blob_uploaded) {
       if (g_vb.blob.ensure(blob_len)) return -2;
-      HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
-                               ctx->stream));
+      if (h2d_staged(ctx, g_vb.blob.p, blob, blob_len, ctx->stream)) return -2;
       HIP_CHECK(hipEventRecord(ctx->ev_blob, ctx->stream));
       blob_uploaded = true;
     }
@@ -1669,6 +1721,7 @@ static int arena_reserve(kv_ctx *ctx, uint64_t need) {
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
   if (ctx->h_s_status) (void)hipHostFree(ctx->h_s_status);
   if (ctx->h_e_status) (void)hipHostFree(ctx->h_e_status);
+  if (ctx->h_stage) (void)hipHostFree(ctx->h_stage);
   if (ctx->d_arena) (void)hipFree(ctx->d_arena);
   ctx->d_arena = na;
   ctx->arena_cap = nc;
