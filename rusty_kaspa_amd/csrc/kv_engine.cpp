@@ -99,6 +99,9 @@ struct kv_ctx {
    * stay valid until the call's final stream syncs. */
   uint8_t *h_stage = nullptr;
   size_t h_stage_cap = 0, h_stage_off = 0;
+  uint8_t *h_mu_partial = nullptr; /* pinned 768B muhash partial landing pad:
+    a pageable-dest D2H of even 384B blocks the enqueue thread until the
+    whole reduce chain drains (measured ~3.9ms) */
   std::vector<uint64_t> found_buf;
   /* GPU-resident UTXO set */
   kv::utxo_slot *d_utxo = nullptr;
@@ -174,6 +177,7 @@ extern "C" void kv_destroy(kv_ctx *ctx) {
   if (ctx->h_s_status) (void)hipHostFree(ctx->h_s_status);
   if (ctx->h_e_status) (void)hipHostFree(ctx->h_e_status);
   if (ctx->h_stage) (void)hipHostFree(ctx->h_stage);
+  if (ctx->h_mu_partial) (void)hipHostFree(ctx->h_mu_partial);
   if (ctx->d_arena) (void)hipFree(ctx->d_arena);
   if (ctx->d_gjobs) (void)hipFree(ctx->d_gjobs);
   if (ctx->d_gout) (void)hipFree(ctx->d_gout);
@@ -874,6 +878,7 @@ static int stage_reserve(kv_ctx *ctx, size_t need) {
   ctx->h_stage_off = 0;
   if (ctx->h_stage_cap >= need) return 0;
   if (ctx->h_stage) (void)hipHostFree(ctx->h_stage);
+  if (ctx->h_mu_partial) (void)hipHostFree(ctx->h_mu_partial);
   size_t nc = need + need / 2 + 4096;
   if (hipHostMalloc(&ctx->h_stage, nc) != hipSuccess) {
     ctx->h_stage = nullptr;
@@ -910,8 +915,15 @@ static inline int h2d_staged(kv_ctx *ctx, void *dst, const void *src, size_t len
  * partial is written synchronously and *launched stays false. */
 static int enqueue_muhash(kv_ctx *ctx, const std::vector<HTx> &txs,
                           const uint8_t *include, uint64_t block_daa_score,
-                          hipStream_t stream, uint8_t *outp, bool *launched,
+                          hipStream_t stream, bool *launched,
                           std::vector<kv::kv_elem_job> &jobs) {
+  if (!ctx->h_mu_partial &&
+      hipHostMalloc(&ctx->h_mu_partial, 768) != hipSuccess) {
+    ctx->h_mu_partial = nullptr;
+    set_error("hipHostMalloc failed (muhash partial)");
+    return -2;
+  }
+  uint8_t *outp = ctx->h_mu_partial;
   jobs.clear();
   for (size_t t = 0; t < txs.size(); t++) {
     if (!include[t]) continue;
@@ -1140,6 +1152,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
   }
 
   auto vt1 = vt_now();
+  auto vt1a = vt1, vt1b = vt1, vt1c = vt1;
 
   /* phase 1.5: sig cache ⇔ TransactionValidator sig_cache (caches.rs:57-82).
    * Key = blake2b-256(tx_id ‖ digest(all input entries) ‖ input_index ‖
@@ -1300,6 +1313,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     }
     HIP_CHECK(hipGetLastError());
   }
+  vt1a = vt_now(); /* phase-2 launches done */
 
   /* optimistic muhash on stream3, overlapped with the verify chains: include
    * every tx that passed the phase-1 integer checks. Phase 4 consumes the
@@ -1319,9 +1333,10 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     }
     HIP_CHECK(hipStreamWaitEvent(ctx->stream3, ctx->ev_blob, 0));
     int mrc = enqueue_muhash(ctx, txs, mu_inc.data(), block_daa_score, ctx->stream3,
-                             muhash_partial_out, &mu_opt_launched, mu_jobs);
+                             &mu_opt_launched, mu_jobs);
     if (mrc) return mrc;
   }
+  vt1b = vt_now(); /* muhash enqueue done */
   /* status readbacks last (pinned, truly async), then join the streams */
   if (ns) {
     if (ctx->h_s_cap < ns) {
@@ -1590,14 +1605,14 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
       for (int t = 0; t < n_txs; t++) mu_inc[t] = codes[t] == 0;
       bool relaunched = false;
       int mrc = enqueue_muhash(ctx, txs, mu_inc.data(), block_daa_score,
-                               ctx->stream, muhash_partial_out, &relaunched,
-                               mu_jobs);
+                               ctx->stream, &relaunched, mu_jobs);
       if (mrc) return mrc;
       HIP_CHECK(hipStreamSynchronize(ctx->stream));
       tev_rec_pair[5] = relaunched;
     } else {
       tev_rec_pair[5] = mu_opt_launched;
     }
+    memcpy(muhash_partial_out, ctx->h_mu_partial, 768);
   }
 
   /* collect per-kernel timings (the stream is synchronized by now) */
@@ -1615,10 +1630,12 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
   memcpy(fees_out, fees.data(), (size_t)n_txs * 8);
   if (kv_timing)
     fprintf(stderr,
-            "[kv_timing] validate: parse+p1 %.2f cache+enq %.2f gpu-wait %.2f "
-            "interp %.2f resolve %.2f muhash-wait %.2f total %.2f ms\n",
-            vt_ms(vt0, vt1), vt_ms(vt1, vt2), vt_ms(vt2, vt3), vt_ms(vt3, vt4),
-            vt_ms(vt4, vt5), vt_ms(vt5, vt_now()), vt_ms(vt0, vt_now()));
+            "[kv_timing] validate: parse+p1 %.2f cache+enq %.2f (p2 %.2f mu %.2f "
+            "rb %.2f) gpu-wait %.2f interp %.2f resolve %.2f muhash-wait %.2f "
+            "total %.2f ms\n",
+            vt_ms(vt0, vt1), vt_ms(vt1, vt2), vt_ms(vt1, vt1a), vt_ms(vt1a, vt1b),
+            vt_ms(vt1b, vt2), vt_ms(vt2, vt3), vt_ms(vt3, vt4), vt_ms(vt4, vt5),
+            vt_ms(vt5, vt_now()), vt_ms(vt0, vt_now()));
   return 0;
 }
 
@@ -1722,6 +1739,7 @@ static int arena_reserve(kv_ctx *ctx, uint64_t need) {
   if (ctx->h_s_status) (void)hipHostFree(ctx->h_s_status);
   if (ctx->h_e_status) (void)hipHostFree(ctx->h_e_status);
   if (ctx->h_stage) (void)hipHostFree(ctx->h_stage);
+  if (ctx->h_mu_partial) (void)hipHostFree(ctx->h_mu_partial);
   if (ctx->d_arena) (void)hipFree(ctx->d_arena);
   ctx->d_arena = na;
   ctx->arena_cap = nc;
