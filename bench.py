@@ -204,7 +204,8 @@ def block_mode(args, dist_ctx=None):
             codes, fees, partial = eng.validate_block_utxo(
                 blob, n_txs, 10**9, 10**9, 0, apply_diff=False)
         else:
-            codes, fees, partial = eng.validate_block(blob, n_txs, 10**9, 10**9, 0)
+            codes, fees, partial = eng.validate_block(blob, n_txs, 10**9, 10**9, 0,
+                                                      raw=True)
         if distributed:
             # the real path's one exchange: allgather the 768B muhash
             # (numerator‖denominator) partials over RCCL, fold the
@@ -221,7 +222,7 @@ def block_mode(args, dist_ctx=None):
             else:
                 mh = None
         else:
-            mh = eng.muhash_finalize(partial)
+            mh = eng.muhash_finalize(bytes(partial))
         return codes, mh
 
     for _ in range(args.warmup):

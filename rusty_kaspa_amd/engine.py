@@ -105,7 +105,7 @@ class Engine:
         self._check(rc)
 
     def validate_block(self, blob: bytes, n_txs: int, pov_daa: int, block_daa: int,
-                       flags: int = 2, want_muhash: bool = True):
+                       flags: int = 2, want_muhash: bool = True, raw: bool = False):
         codes = (ctypes.c_int32 * n_txs)()
         fees = (ctypes.c_uint64 * n_txs)()
         partial = (ctypes.c_uint8 * 768)() if want_muhash else None
@@ -114,6 +114,8 @@ class Engine:
             ctypes.c_uint64(pov_daa), ctypes.c_uint64(block_daa),
             ctypes.c_uint32(flags), codes, fees, partial)
         self._check(rc)
+        if raw:  # bench hot loop: skip the O(n) ctypes->list conversions
+            return codes, fees, partial
         return list(codes), list(fees), bytes(partial) if want_muhash else None
 
     def validate_block_utxo(self, blob: bytes, n_txs: int, pov_daa: int,
