@@ -193,7 +193,15 @@ __device__ inline void b2b_update(b2b_state &S, const uint8_t *data, uint32_t le
     }
     uint32_t take = 128 - S.buflen;
     if (take > len) take = len;
-    for (uint32_t i = 0; i < take; i++) S.buf[S.buflen + i] = data[i];
+    /* copy 8B at a time (unaligned-capable memcpy lowering) — the byte loop
+     * made the subhash kernel walk the blob one byte-load per byte */
+    uint32_t i = 0;
+    for (; i + 8 <= take; i += 8) {
+      uint64_t w;
+      __builtin_memcpy(&w, data + i, 8);
+      __builtin_memcpy(S.buf + S.buflen + i, &w, 8);
+    }
+    for (; i < take; i++) S.buf[S.buflen + i] = data[i];
     S.buflen += take;
     data += take;
     len -= take;

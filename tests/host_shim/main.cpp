@@ -31,15 +31,16 @@ int host_init_gtable() {
   }
   gej acc;
   acc.x = G.x; acc.y = G.y; fe26_set_int(acc.z, 1);
-  for (int k = 1; k <= 15; k++) {
+  KV_G_TABLE8[0] = G;
+  for (int k = 1; k <= 255; k++) {
     fe26 zi, zi2, zi3;
     fe26_inv(zi, acc.z);
     fe26_sqr(zi2, zi);
     fe26_mul(zi3, zi2, zi);
-    fe26_mul(KV_G_TABLE[k].x, acc.x, zi2);
-    fe26_mul(KV_G_TABLE[k].y, acc.y, zi3);
-    fe26_normalize(KV_G_TABLE[k].x);
-    fe26_normalize(KV_G_TABLE[k].y);
+    fe26_mul(KV_G_TABLE8[k].x, acc.x, zi2);
+    fe26_mul(KV_G_TABLE8[k].y, acc.y, zi3);
+    fe26_normalize(KV_G_TABLE8[k].x);
+    fe26_normalize(KV_G_TABLE8[k].y);
     gej t;
     gej_add_ge(t, acc, G);
     acc = t;
