@@ -323,6 +323,86 @@ __device__ KV_GROUP_ATTR void gej_window_step(gej &R, const ge *ptab,
   }
 }
 
+#ifdef KV_PAIR_WINDOWS
+/* Two ladder windows per call frame: 8 doublings + both windows' stream adds
+ * (G digits are 8-bit and land only on the even window). Halves the
+ * accumulator's ABI crossings (17 frames vs 33) at the cost of a ~2x body —
+ * probing the gfx950 long-body hang cliff from below. */
+__device__ KV_GROUP_ATTR void gej_window_step2(gej &R, const ge *ptab,
+                                               const fe26 &beta,
+                                               u64 dg1, u64 ng1, u64 dg2, u64 ng2,
+                                               u64 hp1, u64 hp2, /* odd-window P digits */
+                                               u64 lp1, u64 lp2, /* even-window P digits */
+                                               u64 np1, u64 np2) {
+  gej t;
+  gej_double_impl(t, R);
+  gej_double_impl(R, t);
+  gej_double_impl(t, R);
+  gej_double_impl(R, t);
+  { /* odd window: P streams only */
+    ge e = ptab[hp1];
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)np1);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(hp1 != 0));
+  }
+  {
+    ge e = ptab[hp2];
+    fe26 bx;
+    fe26_mul(bx, e.x, beta);
+    e.x = bx;
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)np2);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(hp2 != 0));
+  }
+  gej_double_impl(t, R);
+  gej_double_impl(R, t);
+  gej_double_impl(t, R);
+  gej_double_impl(R, t);
+  { /* even window: G comb + P streams */
+    ge e = KV_G_TABLE8[dg1];
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)ng1);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(dg1 != 0));
+  }
+  {
+    ge e = KV_G_TABLE8[dg2];
+    fe26 bx;
+    fe26_mul(bx, e.x, beta);
+    e.x = bx;
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)ng2);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(dg2 != 0));
+  }
+  {
+    ge e = ptab[lp1];
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)np1);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(lp1 != 0));
+  }
+  {
+    ge e = ptab[lp2];
+    fe26 bx;
+    fe26_mul(bx, e.x, beta);
+    e.x = bx;
+    fe26 ny;
+    fe26_neg(ny, e.y, 2);
+    fe26_cmov(e.y, ny, (u32)np2);
+    gej_add_ge_impl(t, R, e);
+    gej_cmov(R, t, (u64)(lp2 != 0));
+  }
+}
+#endif
+
 /* R = gs·G + ps·P via GLV-split 4-bit windows: 33 window steps of 4 doublings
  * + 4 selected adds (G, φG, P, φP streams; φ applied at add time as one β·x
  * field multiply; negative half-scalars negate the added point's y). */
@@ -388,6 +468,21 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
     fe26_from_fe(beta, bu);
   }
   gej_set_infinity(R);
+#ifdef KV_PAIR_WINDOWS
+  /* window 32 alone (G digit at even position 32), then 16 window pairs */
+  gej_window_step(R, ptab, beta, 1, glv_digit8(g1h, 32), g1h.neg,
+                  glv_digit8(g2h, 32), g2h.neg, glv_digit(p1h, 32), p1h.neg,
+                  glv_digit(p2h, 32), p2h.neg);
+#pragma unroll 1
+  for (int w = 31; w >= 1; w -= 2) {
+    gej_window_step2(R, ptab, beta,
+                     glv_digit8(g1h, w - 1), g1h.neg,
+                     glv_digit8(g2h, w - 1), g2h.neg,
+                     glv_digit(p1h, w), glv_digit(p2h, w),
+                     glv_digit(p1h, w - 1), glv_digit(p2h, w - 1),
+                     p1h.neg, p2h.neg);
+  }
+#else
 #pragma unroll 1
   for (int w = 32; w >= 0; w--) {
     u64 g_active = (w & 1) == 0; /* 8-bit G digits at even window positions */
@@ -397,6 +492,7 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
                     glv_digit(p1h, w), p1h.neg,
                     glv_digit(p2h, w), p2h.neg);
   }
+#endif
 }
 
 /* One BIP-340 verification; returns KVS_* */
