@@ -323,11 +323,17 @@ __device__ KV_GROUP_ATTR void gej_window_step(gej &R, const ge *ptab,
   }
 }
 
+/* Pair-window frames are the DEFAULT (measured 41.1M vs 39.6M verifies/s);
+ * -DKV_NO_PAIR_WINDOWS restores one-window frames for experiments. */
+#ifndef KV_NO_PAIR_WINDOWS
+#define KV_PAIR_WINDOWS 1
+#endif
+
 #ifdef KV_PAIR_WINDOWS
 /* Two ladder windows per call frame: 8 doublings + both windows' stream adds
  * (G digits are 8-bit and land only on the even window). Halves the
  * accumulator's ABI crossings (17 frames vs 33) at the cost of a ~2x body —
- * probing the gfx950 long-body hang cliff from below. */
+ * under the gfx950 long-body hang cliff (full inlining still hangs). */
 __device__ KV_GROUP_ATTR void gej_window_step2(gej &R, const ge *ptab,
                                                const fe26 &beta,
                                                u64 dg1, u64 ng1, u64 dg2, u64 ng2,
