@@ -334,7 +334,7 @@ __device__ KV_GROUP_ATTR void gej_window_step(gej &R, const ge *ptab,
  * (G digits are 8-bit and land only on the even window). Halves the
  * accumulator's ABI crossings (17 frames vs 33) at the cost of a ~2x body —
  * under the gfx950 long-body hang cliff (full inlining still hangs). */
-__device__ KV_GROUP_ATTR void gej_window_step2(gej &R, const ge *ptab,
+__device__ __forceinline__ void gej_window_pair_impl(gej &R, const ge *ptab,
                                                const fe26 &beta,
                                                u64 dg1, u64 ng1, u64 dg2, u64 ng2,
                                                u64 hp1, u64 hp2, /* odd-window P digits */
@@ -407,6 +407,26 @@ __device__ KV_GROUP_ATTR void gej_window_step2(gej &R, const ge *ptab,
     gej_cmov(R, t, (u64)(lp2 != 0));
   }
 }
+
+__device__ KV_GROUP_ATTR void gej_window_step2(gej &R, const ge *ptab,
+                                               const fe26 &beta, u64 dg1, u64 ng1,
+                                               u64 dg2, u64 ng2, u64 hp1, u64 hp2,
+                                               u64 lp1, u64 lp2, u64 np1, u64 np2) {
+  gej_window_pair_impl(R, ptab, beta, dg1, ng1, dg2, ng2, hp1, hp2, lp1, lp2,
+                       np1, np2);
+}
+
+#ifdef KV_QUAD_WINDOWS
+/* four windows (two pairs) per frame: 9 frames per ladder */
+__device__ KV_GROUP_ATTR void gej_window_step4(gej &R, const ge *ptab,
+                                               const fe26 &beta,
+                                               const u64 a[10], const u64 b[10]) {
+  gej_window_pair_impl(R, ptab, beta, a[0], a[1], a[2], a[3], a[4], a[5], a[6],
+                       a[7], a[8], a[9]);
+  gej_window_pair_impl(R, ptab, beta, b[0], b[1], b[2], b[3], b[4], b[5], b[6],
+                       b[7], b[8], b[9]);
+}
+#endif
 #endif
 
 /* R = gs·G + ps·P via GLV-split 4-bit windows: 33 window steps of 4 doublings
@@ -474,7 +494,24 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
     fe26_from_fe(beta, bu);
   }
   gej_set_infinity(R);
-#ifdef KV_PAIR_WINDOWS
+#if defined(KV_QUAD_WINDOWS)
+  /* window 32 alone, then 8 quad frames (two pairs each) */
+  gej_window_step(R, ptab, beta, 1, glv_digit8(g1h, 32), g1h.neg,
+                  glv_digit8(g2h, 32), g2h.neg, glv_digit(p1h, 32), p1h.neg,
+                  glv_digit(p2h, 32), p2h.neg);
+#pragma unroll 1
+  for (int w = 31; w >= 3; w -= 4) {
+    u64 a[10] = {glv_digit8(g1h, w - 1), g1h.neg, glv_digit8(g2h, w - 1),
+                 g2h.neg, glv_digit(p1h, w), glv_digit(p2h, w),
+                 glv_digit(p1h, w - 1), glv_digit(p2h, w - 1), p1h.neg,
+                 p2h.neg};
+    u64 b[10] = {glv_digit8(g1h, w - 3), g1h.neg, glv_digit8(g2h, w - 3),
+                 g2h.neg, glv_digit(p1h, w - 2), glv_digit(p2h, w - 2),
+                 glv_digit(p1h, w - 3), glv_digit(p2h, w - 3), p1h.neg,
+                 p2h.neg};
+    gej_window_step4(R, ptab, beta, a, b);
+  }
+#elif defined(KV_PAIR_WINDOWS)
   /* window 32 alone (G digit at even position 32), then 16 window pairs */
   gej_window_step(R, ptab, beta, 1, glv_digit8(g1h, 32), g1h.neg,
                   glv_digit8(g2h, 32), g2h.neg, glv_digit(p1h, 32), p1h.neg,
