@@ -65,12 +65,12 @@ ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 545,870
 # gfx950 VALU issue peak: 256 CU x 4 SIMD x 32 lanes x 2.4 GHz = 78.6 T u32/s
 VALU_PEAK_TOPS = 78.6
 # Memory-side traffic per verify, measured by rocprofv3 --pmc FETCH_SIZE /
-# WRITE_SIZE (separate passes) on 262144-tuple staged dispatches of the
-# window-fused kernel: (8.417e6 + 5.626e6) KiB / 262144 = 54,856 B/verify
-# (fetch 32.9KB + write 22.0KB — scratch table reads + the once-per-window
-# accumulator spill; vs 128B of algorithmic input). Provenance:
-# profiles/r02b_verify_window_pmc.json (pre-fusion: 119,296).
-TRAFFIC_BYTES_PER_VERIFY = 54856
+# WRITE_SIZE (separate passes) on 262144-tuple staged dispatches of the FINAL
+# kernel (G comb + pair-window frames): (9.226e6 + 5.536e6) KiB / 262144 =
+# 57,664 B/verify (fetch 36.0KB + write 21.6KB — scratch table reads + the
+# per-frame accumulator spill; vs 128B of algorithmic input). Provenance:
+# profiles/r02c_final_kernel_pmc.json (pre-fusion: 119,296).
+TRAFFIC_BYTES_PER_VERIFY = 57664
 
 
 def log(msg):
