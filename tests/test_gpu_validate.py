@@ -158,3 +158,26 @@ def test_non_template_script_fallback(oracle, engine):
         assert oc[0] == expect0, (oc[0], expect0)
         ec, ef, emh = engine_validate(engine, pb, n)
         assert ec == oc and ef == of and emh == omh
+
+
+def test_validate_timings_exposed(oracle, engine):
+    """kv_get_validate_timings reports per-kernel hipEvent times and job
+    counts of the last validate call (the bench derives the block-path
+    roofline from these)."""
+    class KvTimings(ctypes.Structure):
+        _fields_ = [("subhash_ms", ctypes.c_double),
+                    ("s_assemble_ms", ctypes.c_double),
+                    ("e_assemble_ms", ctypes.c_double),
+                    ("schnorr_ms", ctypes.c_double),
+                    ("ecdsa_ms", ctypes.c_double),
+                    ("muhash_ms", ctypes.c_double),
+                    ("n_schnorr", ctypes.c_uint64),
+                    ("n_ecdsa", ctypes.c_uint64)]
+    blob, _ = gen_block(oracle, seed=99, n_txs=64, pct_ecdsa=20)
+    engine.validate_block(blob, 64, 10**9, 10**9, SKIP_MASS)
+    tm = KvTimings()
+    assert engine.lib.kv_get_validate_timings(ctypes.c_void_p(engine.ctx),
+                                              ctypes.byref(tm)) == 0
+    assert tm.n_schnorr > 0 and tm.n_ecdsa > 0
+    assert tm.subhash_ms > 0 and tm.schnorr_ms > 0 and tm.ecdsa_ms > 0
+    assert tm.muhash_ms > 0  # muhash partial requested by default wrapper
