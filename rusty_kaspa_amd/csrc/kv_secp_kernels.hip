@@ -260,6 +260,21 @@ __device__ __forceinline__ u64 glv_digit8(const glv_half &h, int w) {
   return v & 255;
 }
 
+#ifdef KV_SIGNED_PTAB
+/* signed fixed 4-bit recode: 33 digits in [-8, 8] (LSB order). |h| < 2^130,
+ * so the top digit (<= 4) absorbs the final carry without overflow. */
+__device__ __forceinline__ void glv_recode_signed(const glv_half &h,
+                                                  int8_t dig[33]) {
+  u64 carry = 0;
+#pragma unroll 1
+  for (int w = 0; w < 33; w++) {
+    u64 v = glv_digit(h, w) + carry;
+    carry = v > 8;
+    dig[w] = (int8_t)((long long)v - (long long)(carry << 4));
+  }
+}
+#endif
+
 /* One full ladder window in ONE call frame: 4 doublings + the 4 stream adds
  * (G, φG, P, φP). The group ops inline INSIDE this body, so the accumulator
  * crosses the noinline ABI once per window instead of five times — the r02
@@ -334,12 +349,10 @@ __device__ KV_GROUP_ATTR void gej_window_step(gej &R, const ge *ptab,
  * (G digits are 8-bit and land only on the even window). Halves the
  * accumulator's ABI crossings (17 frames vs 33) at the cost of a ~2x body —
  * under the gfx950 long-body hang cliff (full inlining still hangs). */
-__device__ __forceinline__ void gej_window_pair_impl(gej &R, const ge *ptab,
-                                               const fe26 &beta,
-                                               u64 dg1, u64 ng1, u64 dg2, u64 ng2,
-                                               u64 hp1, u64 hp2, /* odd-window P digits */
-                                               u64 lp1, u64 lp2, /* even-window P digits */
-                                               u64 np1, u64 np2) {
+__device__ __forceinline__ void gej_window_pair_impl(
+    gej &R, const ge *ptab, const fe26 &beta, u64 dg1, u64 ng1, u64 dg2, u64 ng2,
+    u64 hp1, u64 hn1, u64 hp2, u64 hn2, /* odd-window P digits + negs */
+    u64 lp1, u64 ln1, u64 lp2, u64 ln2 /* even-window P digits + negs */) {
   gej t;
   gej_double_impl(t, R);
   gej_double_impl(R, t);
@@ -349,7 +362,7 @@ __device__ __forceinline__ void gej_window_pair_impl(gej &R, const ge *ptab,
     ge e = ptab[hp1];
     fe26 ny;
     fe26_neg(ny, e.y, 2);
-    fe26_cmov(e.y, ny, (u32)np1);
+    fe26_cmov(e.y, ny, (u32)hn1);
     gej_add_ge_impl(t, R, e);
     gej_cmov(R, t, (u64)(hp1 != 0));
   }
@@ -360,7 +373,7 @@ __device__ __forceinline__ void gej_window_pair_impl(gej &R, const ge *ptab,
     e.x = bx;
     fe26 ny;
     fe26_neg(ny, e.y, 2);
-    fe26_cmov(e.y, ny, (u32)np2);
+    fe26_cmov(e.y, ny, (u32)hn2);
     gej_add_ge_impl(t, R, e);
     gej_cmov(R, t, (u64)(hp2 != 0));
   }
@@ -391,7 +404,7 @@ __device__ __forceinline__ void gej_window_pair_impl(gej &R, const ge *ptab,
     ge e = ptab[lp1];
     fe26 ny;
     fe26_neg(ny, e.y, 2);
-    fe26_cmov(e.y, ny, (u32)np1);
+    fe26_cmov(e.y, ny, (u32)ln1);
     gej_add_ge_impl(t, R, e);
     gej_cmov(R, t, (u64)(lp1 != 0));
   }
@@ -402,7 +415,7 @@ __device__ __forceinline__ void gej_window_pair_impl(gej &R, const ge *ptab,
     e.x = bx;
     fe26 ny;
     fe26_neg(ny, e.y, 2);
-    fe26_cmov(e.y, ny, (u32)np2);
+    fe26_cmov(e.y, ny, (u32)ln2);
     gej_add_ge_impl(t, R, e);
     gej_cmov(R, t, (u64)(lp2 != 0));
   }
@@ -410,21 +423,22 @@ __device__ __forceinline__ void gej_window_pair_impl(gej &R, const ge *ptab,
 
 __device__ KV_GROUP_ATTR void gej_window_step2(gej &R, const ge *ptab,
                                                const fe26 &beta, u64 dg1, u64 ng1,
-                                               u64 dg2, u64 ng2, u64 hp1, u64 hp2,
-                                               u64 lp1, u64 lp2, u64 np1, u64 np2) {
-  gej_window_pair_impl(R, ptab, beta, dg1, ng1, dg2, ng2, hp1, hp2, lp1, lp2,
-                       np1, np2);
+                                               u64 dg2, u64 ng2, u64 hp1, u64 hn1,
+                                               u64 hp2, u64 hn2, u64 lp1, u64 ln1,
+                                               u64 lp2, u64 ln2) {
+  gej_window_pair_impl(R, ptab, beta, dg1, ng1, dg2, ng2, hp1, hn1, hp2, hn2,
+                       lp1, ln1, lp2, ln2);
 }
 
 #ifdef KV_QUAD_WINDOWS
 /* four windows (two pairs) per frame: 9 frames per ladder */
 __device__ KV_GROUP_ATTR void gej_window_step4(gej &R, const ge *ptab,
                                                const fe26 &beta,
-                                               const u64 a[10], const u64 b[10]) {
+                                               const u64 a[12], const u64 b[12]) {
   gej_window_pair_impl(R, ptab, beta, a[0], a[1], a[2], a[3], a[4], a[5], a[6],
-                       a[7], a[8], a[9]);
+                       a[7], a[8], a[9], a[10], a[11]);
   gej_window_pair_impl(R, ptab, beta, b[0], b[1], b[2], b[3], b[4], b[5], b[6],
-                       b[7], b[8], b[9]);
+                       b[7], b[8], b[9], b[10], b[11]);
 }
 #endif
 #endif
@@ -445,16 +459,21 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
    * MALL), and the P streams become MIXED adds (11 vs 16 fe_muls); the one
    * extra field inversion (+~360 fe_muls with the back-substitution) is
    * paid back by the 66 cheaper ladder adds (-330). */
-  ge ptab[16];
+#ifdef KV_SIGNED_PTAB
+#define KV_PTAB_MAX 8
+#else
+#define KV_PTAB_MAX 15
+#endif
+  ge ptab[KV_PTAB_MAX + 1];
   {
-    fe26 ztab[16], pref[16];
+    fe26 ztab[KV_PTAB_MAX + 1], pref[KV_PTAB_MAX + 1];
     gej acc;
     acc.x = P.x;
     acc.y = P.y;
     fe26_set_int(acc.z, 1);
     ptab[1] = P;
 #pragma unroll 1
-    for (int k = 2; k <= 15; k++) {
+    for (int k = 2; k <= KV_PTAB_MAX; k++) {
       gej t;
       gej_add_ge(t, acc, P);
       acc = t;
@@ -462,14 +481,14 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
       ptab[k].y = acc.y;
       ztab[k] = acc.z;
     }
-    /* batch-invert z2..z15: one fe26_inv + 3 muls per entry */
+    /* batch-invert z2..zmax: one fe26_inv + 3 muls per entry */
     pref[2] = ztab[2];
 #pragma unroll 1
-    for (int k = 3; k <= 15; k++) fe26_mul(pref[k], pref[k - 1], ztab[k]);
+    for (int k = 3; k <= KV_PTAB_MAX; k++) fe26_mul(pref[k], pref[k - 1], ztab[k]);
     fe26 inv;
-    fe26_inv(inv, pref[15]);
+    fe26_inv(inv, pref[KV_PTAB_MAX]);
 #pragma unroll 1
-    for (int k = 15; k >= 2; k--) {
+    for (int k = KV_PTAB_MAX; k >= 2; k--) {
       fe26 zi;
       if (k > 2) {
         fe26_mul(zi, inv, pref[k - 1]);
@@ -486,6 +505,11 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
     ptab[0] = ptab[1]; /* digit 0: add computed then discarded by cmov — a
                           well-formed point keeps magnitudes in range */
   }
+#ifdef KV_SIGNED_PTAB
+  int8_t dp1[33], dp2[33];
+  glv_recode_signed(p1h, dp1);
+  glv_recode_signed(p2h, dp2);
+#endif
   fe26 beta;
   {
     fe bu;
@@ -494,36 +518,46 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
     fe26_from_fe(beta, bu);
   }
   gej_set_infinity(R);
+  /* per-stream digit/neg accessors: unsigned fixed windows by default,
+   * signed fixed windows (8-entry table) under KV_SIGNED_PTAB */
+#ifdef KV_SIGNED_PTAB
+#define KV_DP1(w) ((u64)(dp1[w] < 0 ? -dp1[w] : dp1[w]))
+#define KV_DN1(w) ((u64)((dp1[w] < 0) ^ (int)p1h.neg))
+#define KV_DP2(w) ((u64)(dp2[w] < 0 ? -dp2[w] : dp2[w]))
+#define KV_DN2(w) ((u64)((dp2[w] < 0) ^ (int)p2h.neg))
+#else
+#define KV_DP1(w) glv_digit(p1h, w)
+#define KV_DN1(w) (p1h.neg)
+#define KV_DP2(w) glv_digit(p2h, w)
+#define KV_DN2(w) (p2h.neg)
+#endif
 #if defined(KV_QUAD_WINDOWS)
   /* window 32 alone, then 8 quad frames (two pairs each) */
   gej_window_step(R, ptab, beta, 1, glv_digit8(g1h, 32), g1h.neg,
-                  glv_digit8(g2h, 32), g2h.neg, glv_digit(p1h, 32), p1h.neg,
-                  glv_digit(p2h, 32), p2h.neg);
+                  glv_digit8(g2h, 32), g2h.neg, KV_DP1(32), KV_DN1(32),
+                  KV_DP2(32), KV_DN2(32));
 #pragma unroll 1
   for (int w = 31; w >= 3; w -= 4) {
-    u64 a[10] = {glv_digit8(g1h, w - 1), g1h.neg, glv_digit8(g2h, w - 1),
-                 g2h.neg, glv_digit(p1h, w), glv_digit(p2h, w),
-                 glv_digit(p1h, w - 1), glv_digit(p2h, w - 1), p1h.neg,
-                 p2h.neg};
-    u64 b[10] = {glv_digit8(g1h, w - 3), g1h.neg, glv_digit8(g2h, w - 3),
-                 g2h.neg, glv_digit(p1h, w - 2), glv_digit(p2h, w - 2),
-                 glv_digit(p1h, w - 3), glv_digit(p2h, w - 3), p1h.neg,
-                 p2h.neg};
+    u64 a[12] = {glv_digit8(g1h, w - 1), g1h.neg, glv_digit8(g2h, w - 1),
+                 g2h.neg, KV_DP1(w), KV_DN1(w), KV_DP2(w), KV_DN2(w),
+                 KV_DP1(w - 1), KV_DN1(w - 1), KV_DP2(w - 1), KV_DN2(w - 1)};
+    u64 b[12] = {glv_digit8(g1h, w - 3), g1h.neg, glv_digit8(g2h, w - 3),
+                 g2h.neg, KV_DP1(w - 2), KV_DN1(w - 2), KV_DP2(w - 2), KV_DN2(w - 2),
+                 KV_DP1(w - 3), KV_DN1(w - 3), KV_DP2(w - 3), KV_DN2(w - 3)};
     gej_window_step4(R, ptab, beta, a, b);
   }
 #elif defined(KV_PAIR_WINDOWS)
   /* window 32 alone (G digit at even position 32), then 16 window pairs */
   gej_window_step(R, ptab, beta, 1, glv_digit8(g1h, 32), g1h.neg,
-                  glv_digit8(g2h, 32), g2h.neg, glv_digit(p1h, 32), p1h.neg,
-                  glv_digit(p2h, 32), p2h.neg);
+                  glv_digit8(g2h, 32), g2h.neg, KV_DP1(32), KV_DN1(32),
+                  KV_DP2(32), KV_DN2(32));
 #pragma unroll 1
   for (int w = 31; w >= 1; w -= 2) {
     gej_window_step2(R, ptab, beta,
                      glv_digit8(g1h, w - 1), g1h.neg,
                      glv_digit8(g2h, w - 1), g2h.neg,
-                     glv_digit(p1h, w), glv_digit(p2h, w),
-                     glv_digit(p1h, w - 1), glv_digit(p2h, w - 1),
-                     p1h.neg, p2h.neg);
+                     KV_DP1(w), KV_DN1(w), KV_DP2(w), KV_DN2(w),
+                     KV_DP1(w - 1), KV_DN1(w - 1), KV_DP2(w - 1), KV_DN2(w - 1));
   }
 #else
 #pragma unroll 1
@@ -532,10 +566,14 @@ __device__ inline void ecmult_double(gej &R, const sc &gs, const sc &ps,
     gej_window_step(R, ptab, beta, g_active,
                     g_active ? glv_digit8(g1h, w) : 0, g1h.neg,
                     g_active ? glv_digit8(g2h, w) : 0, g2h.neg,
-                    glv_digit(p1h, w), p1h.neg,
-                    glv_digit(p2h, w), p2h.neg);
+                    KV_DP1(w), KV_DN1(w), KV_DP2(w), KV_DN2(w));
   }
 #endif
+#undef KV_DP1
+#undef KV_DN1
+#undef KV_DP2
+#undef KV_DN2
+#undef KV_PTAB_MAX
 }
 
 /* One BIP-340 verification; returns KVS_* */
