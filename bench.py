@@ -49,19 +49,19 @@ DATA_DIR = os.path.join(REPO, "bench_data")
 # comb digits for the shared-G streams — does
 #   132 Jacobian doubles  x  7 fe_mul-equiv      =  924
 #   100 mixed adds        x 11                   = 1100
-#     (66 P-stream + 34 G-stream comb adds)
+#     (66 signed-window P-stream + 34 G-comb)
 #    34 phi-multiplies    x  1                   =   34
-#    15-entry P-table build 14 x 11              =  154
-#    table batch-inversion (prefix 13 + inv 269
-#      + back-sub 26 + 4x14 per-entry)           =  364
+#     8-entry signed P-table build 7 x 11        =   77
+#    table batch-inversion (prefix 6 + inv 269
+#      + back-sub 12 + 4x7 per-entry)            =  315
 #    x-lift sqrt (addition chain)                =  266
 #    final inversion (addition chain)            =  269  (+ ~100 misc: scalar
-#    decomposition muls, challenge/normalize)    ≈ 3,211 fe_mul-equivalents
+#    decomposition muls, digit recode, challenge) ≈ 3,085 fe_mul-equivalents
 # per verify; each 10x26-limb fe_mul ≈ 170 u32-ALU-op equivalents (100 v_mad
-# 32x32 column products + fold/normalize) → ≈ 0.55e6 u32-ops per verify.
-ALG_FE_MULS_PER_VERIFY = 3211
+# 32x32 column products + fold/normalize) → ≈ 0.52e6 u32-ops per verify.
+ALG_FE_MULS_PER_VERIFY = 3085
 ALG_OPS_PER_FE_MUL = 170
-ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 545,870
+ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 524,450
 # gfx950 VALU issue peak: 256 CU x 4 SIMD x 32 lanes x 2.4 GHz = 78.6 T u32/s
 VALU_PEAK_TOPS = 78.6
 # Memory-side traffic per verify, measured by rocprofv3 --pmc FETCH_SIZE /
