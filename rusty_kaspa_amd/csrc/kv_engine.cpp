@@ -33,13 +33,19 @@ static thread_local std::string g_last_error;
 
 static void set_error(const char *msg) { g_last_error = msg ? msg : ""; }
 
+static void set_error_loc(const char *msg, const char *file, int line,
+                          const char *expr) {
+  g_last_error = std::string(msg ? msg : "") + " at " + file + ":" +
+                 std::to_string(line) + " (" + expr + ")";
+}
+
 extern "C" const char *kv_last_error(void) { return g_last_error.c_str(); }
 
 #define HIP_CHECK(expr)                                                        \
   do {                                                                         \
     hipError_t err_ = (expr);                                                  \
     if (err_ != hipSuccess) {                                                  \
-      set_error(hipGetErrorString(err_));                                      \
+      set_error_loc(hipGetErrorString(err_), __FILE__, __LINE__, #expr);       \
       return -2;                                                               \
     }                                                                          \
   } while (0)
@@ -186,6 +192,16 @@ extern "C" void kv_destroy(kv_ctx *ctx) {
   (void)hipStreamDestroy(ctx->stream3);
   (void)hipEventDestroy(ctx->ev_blob);
   (void)hipEventDestroy(ctx->ev_sub);
+  if (ctx->tev_init)
+    for (int i = 0; i < 12; i++) (void)hipEventDestroy(ctx->tev[i]);
+  if (ctx->d_utxo) (void)hipFree(ctx->d_utxo);
+  if (ctx->d_op_in) (void)hipFree(ctx->d_op_in);
+  if (ctx->d_val_in) (void)hipFree(ctx->d_val_in);
+  if (ctx->d_ent_out) (void)hipFree(ctx->d_ent_out);
+  (void)hipGetLastError(); /* teardown calls are fire-and-forget; do not leave
+      their errors sticky for the next engine's launch checks (a swallowed
+      invalid-argument here surfaced as a spurious failure in the NEXT
+      context's first kernel-launch check) */
   delete ctx;
 }
 
@@ -195,6 +211,8 @@ static int verify_batch(kv_ctx *ctx, const uint8_t *tuples, size_t n, size_t rec
                         int ecdsa, uint64_t *bitmap_out, uint8_t *status_out) {
   if (n == 0) return 0;
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   size_t words = (n + 63) / 64;
   if (ensure_cap((void **)&ctx->d_in, &ctx->d_in_cap, n * rec_size)) return -2;
   if (ensure_cap((void **)&ctx->d_bitmap, &ctx->d_bitmap_cap, words * 8)) return -2;
@@ -382,6 +400,8 @@ extern "C" int kv_muhash_finalize(kv_ctx *ctx, const uint8_t *partial768,
 extern "C" int kv_set_seq_commit_accessor(kv_ctx *ctx, kv_seq_commit_accessor_fn fn,
                                           void *user) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   ctx->seqc_fn = fn;
   ctx->seqc_user = user;
   return 0;
@@ -390,6 +410,8 @@ extern "C" int kv_set_seq_commit_accessor(kv_ctx *ctx, kv_seq_commit_accessor_fn
 extern "C" int kv_get_validate_timings(kv_ctx *ctx, kv_validate_timings *out) {
   if (!ctx || !out) return -1;
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   *out = ctx->last_timings;
   return 0;
 }
@@ -411,6 +433,8 @@ extern "C" int kv_sig_cache_stats(kv_ctx *ctx, kv_cache_stats *out) {
 extern "C" int kv_stage_tuples(kv_ctx *ctx, const uint8_t *tuples, size_t n,
                                int ecdsa) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   size_t rec = ecdsa ? 132 : 128;
   size_t words = (n + 63) / 64;
   if (ensure_cap((void **)&ctx->d_in, &ctx->d_in_cap, n * rec)) return -2;
@@ -423,6 +447,8 @@ extern "C" int kv_stage_tuples(kv_ctx *ctx, const uint8_t *tuples, size_t n,
  * Does NOT copy the bitmap back (kv_fetch_bitmap does). */
 extern "C" int kv_verify_staged(kv_ctx *ctx, size_t n, int ecdsa, double *kernel_ms) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   int block = 256;
   unsigned long long grid = (n + block - 1) / block;
   hipEvent_t t0, t1;
@@ -450,6 +476,8 @@ extern "C" int kv_verify_staged(kv_ctx *ctx, size_t n, int ecdsa, double *kernel
 
 extern "C" int kv_fetch_bitmap(kv_ctx *ctx, size_t n, uint64_t *bitmap_out) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   size_t words = (n + 63) / 64;
   HIP_CHECK(hipMemcpy(bitmap_out, ctx->d_bitmap, words * 8, hipMemcpyDeviceToHost));
   return 0;
@@ -780,6 +808,8 @@ extern "C" int kv_sighash_batch(kv_ctx *ctx, const uint8_t *blob, size_t blob_le
    * kind — the assemble kernel uses different tuple strides for schnorr (128B)
    * and ecdsa (132B). */
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   vector<HTx> txs;
   if (parse_blob_host(blob, blob_len, txs) < 0) {
     set_error("kv_sighash_batch: malformed blob");
@@ -1648,6 +1678,8 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
     return -1;
   }
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   return validate_block_impl(ctx, blob, blob_len, pov_daa_score, block_daa_score,
                              flags, tx_codes_out, fees_out, muhash_partial_out,
                              nullptr);
@@ -1662,6 +1694,8 @@ extern "C" int kv_validate_block(kv_ctx *ctx, const uint8_t *blob, size_t blob_l
 
 extern "C" int kv_utxo_reset(kv_ctx *ctx, uint64_t capacity) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   uint64_t cap = 64;
   while (cap < capacity * 2) cap <<= 1; /* ≤50% load factor */
   if (ctx->d_utxo) (void)hipFree(ctx->d_utxo);
@@ -1718,6 +1752,8 @@ static int utxo_upsert_nolock(kv_ctx *ctx, const uint8_t *outpoints,
 extern "C" int kv_utxo_upsert(kv_ctx *ctx, const uint8_t *outpoints,
                               const uint8_t *entries64, size_t n) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   return utxo_upsert_nolock(ctx, outpoints, entries64, n);
 }
 
@@ -1803,6 +1839,8 @@ extern "C" int kv_utxo_upsert_spk(kv_ctx *ctx, const uint8_t *outpoints,
                                   const uint8_t *entries64, const uint8_t *spk_blob,
                                   size_t spk_blob_len, size_t n) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   return utxo_upsert_spk_nolock(ctx, outpoints, entries64, spk_blob,
                                 spk_blob_len, n);
 }
@@ -1824,6 +1862,8 @@ static int utxo_remove_nolock(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
 
 extern "C" int kv_utxo_remove(kv_ctx *ctx, const uint8_t *outpoints, size_t n) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   return utxo_remove_nolock(ctx, outpoints, n);
 }
 
@@ -1877,6 +1917,8 @@ extern "C" int kv_utxo_lookup_spk(kv_ctx *ctx, const uint8_t *outpoints, size_t 
                                   uint8_t *spk_out, size_t spk_cap,
                                   size_t *spk_used) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   int rc = utxo_lookup_nolock(ctx, outpoints, n, entries_out, found_bitmap,
                               nullptr);
   if (rc) return rc;
@@ -1951,6 +1993,8 @@ extern "C" int kv_utxo_lookup(kv_ctx *ctx, const uint8_t *outpoints, size_t n,
                               uint8_t *entries_out, uint64_t *found_bitmap,
                               double *kernel_ms) {
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   return utxo_lookup_nolock(ctx, outpoints, n, entries_out, found_bitmap,
                             kernel_ms);
 }
@@ -1980,6 +2024,8 @@ extern "C" int kv_validate_mempool(kv_ctx *ctx, const uint8_t *blob,
     return -1;
   }
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   vector<HTx> txs;
   int n_txs = parse_blob_host(blob, blob_len, txs);
   if (n_txs < 0) {
@@ -2047,6 +2093,8 @@ extern "C" int kv_block_body_check(kv_ctx *ctx, const uint8_t *blob,
     return -1;
   }
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   vector<HTx> txs;
   int n_txs = parse_blob_host(blob, blob_len, txs);
   if (n_txs < 0) {
@@ -2321,6 +2369,8 @@ extern "C" int kv_validate_block_utxo(kv_ctx *ctx, const uint8_t *blob,
     return -1;
   }
   std::lock_guard<std::mutex> lk(ctx->mu);
+  (void)hipGetLastError(); /* clear any stale per-thread error so the
+      launch-config checks below only see THIS call's launches */
   if (!ctx->d_utxo) {
     set_error("kv_validate_block_utxo: call kv_utxo_reset first");
     return -1;
