@@ -38,6 +38,12 @@ def _worker(rank, world, port, results):
     bitmap[rank] = -1  # all 64 bits set
     dist.all_reduce(bitmap, op=dist.ReduceOp.SUM)  # disjoint shards: SUM == OR
 
+    # the bench's verify leg runs IDENTICAL cached batches per rank and
+    # exchanges with MAX (bit-exact when every rank's words agree)
+    bitmap_max = torch.full((world,), 0x0F0F, dtype=torch.int64)
+    dist.all_reduce(bitmap_max, op=dist.ReduceOp.MAX)
+    assert (bitmap_max == 0x0F0F).all()
+
     # muhash partial exchange: all-gather + local multiplicative fold
     t = torch.frombuffer(bytes(partial), dtype=torch.uint8).clone()
     gathered = [torch.zeros_like(t) for _ in range(world)]
