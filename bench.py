@@ -400,6 +400,10 @@ def verify_mode(args):
     log(f"verdict bitmap spot-check vs oracle OK ({sample} tuples)")
 
     avg_kernel_ms = sum(kernel_times) / len(kernel_times)
+    log(f"kernel ms/step min={min(kernel_times):.2f} "
+        f"avg={avg_kernel_ms:.2f} max={max(kernel_times):.2f} "
+        f"(spread > ~10% = the box is clock-throttling; box-to-box spread "
+        f"20-45M verifies/s observed on this pool)")
     achieved_tops = ALG_OPS_PER_VERIFY * n / (avg_kernel_ms / 1e3) / 1e12
     roofline = {
         "bound": "valu",  # integer-VALU compute roofline: no MFMA path exists
