@@ -266,3 +266,39 @@ def test_zk_defers_with_distinct_code(oracle, engine):
     ecodes, _, _ = engine.validate_block(blob, 1, 10**9, 10**9, SKIP_MASS)
     assert ocodes[0] % 100 == ORACLE_UNSUPPORTED
     assert ecodes[0] == DEFER
+
+
+def test_forty_sites_forty_rounds(oracle, engine):
+    """Depth stress: 40 sequential checksig sites (one GPU round each),
+    verdicts parked on the alt stack and AND-folded — all real signatures."""
+    key, pk = keypair(oracle, 29)
+    n_sites = 40
+    spk = (push(pk) + b"\xac\x6b") * n_sites + b"\x6c" * n_sites + \
+          b"\x9a" * (n_sites - 1)
+    placeholder = push(bytes(65)) * n_sites
+    funding_seq = spend_tx(oracle, placeholder, spk)
+    # raise the compute-budget commitment: 40 sigops > the default 20
+    ins = list(funding_seq["inputs"])
+    ins[0] = dict(ins[0], commit_value=50)
+    funding_seq = dict(funding_seq, inputs=ins)
+    txs = [funding_seq]
+    blob = B.build_blob(txs)
+    msg = (ctypes.c_uint8 * 32)()
+    assert oracle.ok_sighash(blob, len(blob), 0, 0, 1, 0, msg) == 0
+    sig = (ctypes.c_uint8 * 64)()
+    assert oracle.ok_schnorr_sign(key, bytes(msg), None, sig) == 1
+    sp = push(bytes(sig) + b"\x01")
+    ins = list(txs[0]["inputs"])
+    ins[0] = dict(ins[0], sig_script=sp * n_sites)
+    txs[0] = dict(txs[0], inputs=ins)
+    ocodes, ecodes = validate_both(oracle, engine, txs)
+    assert ecodes == ocodes == [0], (ocodes, ecodes)
+    # flip one middle signature: both sides reject identically
+    bad = bytearray(bytes(sig))
+    bad[7] ^= 2
+    scripts = [sp] * n_sites
+    scripts[20] = push(bytes(bad) + b"\x01")
+    ins[0] = dict(ins[0], sig_script=b"".join(scripts))
+    txs[0] = dict(txs[0], inputs=ins)
+    ocodes, ecodes = validate_both(oracle, engine, txs)
+    assert ecodes == ocodes and ocodes[0] != 0
