@@ -97,6 +97,8 @@ __device__ __constant__ static const u64 GE_G_Y[4] = {
  * (17 digits per 129-bit half-scalar instead of 33) — 32 fewer mixed adds
  * per verify. */
 __device__ ge KV_G_TABLE8[256];
+__device__ static fe26 KV_G8_Z[256];    /* init-time scratch */
+__device__ static fe26 KV_G8_PREF[256]; /* init-time scratch */
 
 extern "C" __global__ void kv_ec_table_init_kernel() {
   if (threadIdx.x != 0 || blockIdx.x != 0) return;
@@ -111,25 +113,41 @@ extern "C" __global__ void kv_ec_table_init_kernel() {
     fe26_from_fe(G.x, gx);
     fe26_from_fe(G.y, gy);
   }
+  /* one Montgomery batch inversion over all 255 Zs instead of 255 Fermat
+   * chains (the serial per-entry form cost ~35ms of every kv_create) */
   gej acc;
   acc.x = G.x;
   acc.y = G.y;
   fe26_set_int(acc.z, 1);
   KV_G_TABLE8[0] = G;
   for (int k = 1; k <= 255; k++) {
-    fe26 zi, zi2, zi3;
-    fe26_inv(zi, acc.z);
-    fe26_sqr(zi2, zi);
-    fe26_mul(zi3, zi2, zi);
-    ge e;
-    fe26_mul(e.x, acc.x, zi2);
-    fe26_mul(e.y, acc.y, zi3);
-    fe26_normalize(e.x);
-    fe26_normalize(e.y);
-    KV_G_TABLE8[k] = e;
+    KV_G_TABLE8[k].x = acc.x;
+    KV_G_TABLE8[k].y = acc.y;
+    KV_G8_Z[k] = acc.z;
     gej t;
     gej_add_ge(t, acc, G);
     acc = t;
+  }
+  KV_G8_PREF[1] = KV_G8_Z[1];
+  for (int k = 2; k <= 255; k++)
+    fe26_mul(KV_G8_PREF[k], KV_G8_PREF[k - 1], KV_G8_Z[k]);
+  fe26 inv;
+  fe26_inv(inv, KV_G8_PREF[255]);
+  for (int k = 255; k >= 1; k--) {
+    fe26 zi;
+    if (k > 1) {
+      fe26_mul(zi, inv, KV_G8_PREF[k - 1]);
+      fe26_mul(inv, inv, KV_G8_Z[k]);
+    } else {
+      zi = inv;
+    }
+    fe26 zi2, zi3;
+    fe26_sqr(zi2, zi);
+    fe26_mul(zi3, zi2, zi);
+    fe26_mul(KV_G_TABLE8[k].x, KV_G_TABLE8[k].x, zi2);
+    fe26_mul(KV_G_TABLE8[k].y, KV_G_TABLE8[k].y, zi3);
+    fe26_normalize(KV_G_TABLE8[k].x);
+    fe26_normalize(KV_G_TABLE8[k].y);
   }
 }
 

@@ -32,18 +32,30 @@ int host_init_gtable() {
   gej acc;
   acc.x = G.x; acc.y = G.y; fe26_set_int(acc.z, 1);
   KV_G_TABLE8[0] = G;
+  static fe26 zs[256], pref[256];
   for (int k = 1; k <= 255; k++) {
-    fe26 zi, zi2, zi3;
-    fe26_inv(zi, acc.z);
-    fe26_sqr(zi2, zi);
-    fe26_mul(zi3, zi2, zi);
-    fe26_mul(KV_G_TABLE8[k].x, acc.x, zi2);
-    fe26_mul(KV_G_TABLE8[k].y, acc.y, zi3);
-    fe26_normalize(KV_G_TABLE8[k].x);
-    fe26_normalize(KV_G_TABLE8[k].y);
+    KV_G_TABLE8[k].x = acc.x;
+    KV_G_TABLE8[k].y = acc.y;
+    zs[k] = acc.z;
     gej t;
     gej_add_ge(t, acc, G);
     acc = t;
+  }
+  pref[1] = zs[1];
+  for (int k = 2; k <= 255; k++) fe26_mul(pref[k], pref[k - 1], zs[k]);
+  fe26 inv;
+  fe26_inv(inv, pref[255]);
+  for (int k = 255; k >= 1; k--) {
+    fe26 zi;
+    if (k > 1) { fe26_mul(zi, inv, pref[k - 1]); fe26_mul(inv, inv, zs[k]); }
+    else zi = inv;
+    fe26 zi2, zi3;
+    fe26_sqr(zi2, zi);
+    fe26_mul(zi3, zi2, zi);
+    fe26_mul(KV_G_TABLE8[k].x, KV_G_TABLE8[k].x, zi2);
+    fe26_mul(KV_G_TABLE8[k].y, KV_G_TABLE8[k].y, zi3);
+    fe26_normalize(KV_G_TABLE8[k].x);
+    fe26_normalize(KV_G_TABLE8[k].y);
   }
   return 1;
 }
