@@ -72,6 +72,10 @@ struct kv_ctx {
   /* KIP-21 seq-commitment accessor (kv_set_seq_commit_accessor) */
   kv_seq_commit_accessor_fn seqc_fn = nullptr;
   void *seqc_user = nullptr;
+  /* per-context validate scratch (ValidateBufs; opaque here because the type
+   * lives with the validate section). Per-ctx so two engine contexts in one
+   * process never share device scratch. */
+  void *vb = nullptr;
   /* per-kernel timing events of the last validate call (kv_get_validate_timings):
    * pairs (start,stop) for subhash, s-assemble, e-assemble, schnorr, ecdsa, muhash */
   hipEvent_t tev[12] = {};
@@ -175,8 +179,13 @@ extern "C" kv_ctx *kv_create(const kv_params *params) {
   return ctx;
 }
 
+namespace {
+void kv_vb_release(void *vb); /* defined with ValidateBufs below */
+}
+
 extern "C" void kv_destroy(kv_ctx *ctx) {
   if (!ctx) return;
+  if (ctx->vb) kv_vb_release(ctx->vb);
   if (ctx->d_in) (void)hipFree(ctx->d_in);
   if (ctx->d_bitmap) (void)hipFree(ctx->d_bitmap);
   if (ctx->d_status) (void)hipFree(ctx->d_status);
@@ -514,9 +523,29 @@ struct DeviceBuf {
 struct ValidateBufs {
   DeviceBuf blob, subhashes, s_jobs, e_jobs, s_tuples, e_tuples, s_bitmap, e_bitmap,
       s_status, e_status, elem_jobs, elements, partials_a, partials_b, tx_hashes;
+  void release() {
+    DeviceBuf *bufs[] = {&blob, &subhashes, &s_jobs, &e_jobs, &s_tuples,
+                         &e_tuples, &s_bitmap, &e_bitmap, &s_status, &e_status,
+                         &elem_jobs, &elements, &partials_a, &partials_b,
+                         &tx_hashes};
+    for (auto *b : bufs) {
+      if (b->p) (void)hipFree(b->p);
+      b->p = nullptr;
+      b->cap = 0;
+    }
+  }
 };
 
-static ValidateBufs g_vb; /* guarded by ctx->mu (single validate at a time) */
+static ValidateBufs &vb_of(kv_ctx *ctx) {
+  if (!ctx->vb) ctx->vb = new ValidateBufs();
+  return *(ValidateBufs *)ctx->vb;
+}
+
+void kv_vb_release(void *p) {
+  auto *vb = (ValidateBufs *)p;
+  vb->release();
+  delete vb;
+}
 
 /* classify one input; appends jobs. Returns plan. */
 /* non-template scripts: run the host general interpreter (kv_script_host.inc)
@@ -808,6 +837,7 @@ extern "C" int kv_sighash_batch(kv_ctx *ctx, const uint8_t *blob, size_t blob_le
    * kind — the assemble kernel uses different tuple strides for schnorr (128B)
    * and ecdsa (132B). */
   std::lock_guard<std::mutex> lk(ctx->mu);
+  ValidateBufs &vb = vb_of(ctx);
   (void)hipGetLastError(); /* clear any stale per-thread error so the
       launch-config checks below only see THIS call's launches */
   vector<HTx> txs;
@@ -834,44 +864,44 @@ extern "C" int kv_sighash_batch(kv_ctx *ctx, const uint8_t *blob, size_t blob_le
     }
   }
   uint32_t n_txs = (uint32_t)txs.size();
-  if (g_vb.blob.ensure(blob_len) || g_vb.subhashes.ensure((size_t)n_txs * 160))
+  if (vb.blob.ensure(blob_len) || vb.subhashes.ensure((size_t)n_txs * 160))
     return -2;
-  HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
+  HIP_CHECK(hipMemcpyAsync(vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
                            ctx->stream));
   hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256), dim3(256), 0,
-                     ctx->stream, (const uint8_t *)g_vb.blob.p, n_txs,
-                     (uint8_t *)g_vb.subhashes.p);
+                     ctx->stream, (const uint8_t *)vb.blob.p, n_txs,
+                     (uint8_t *)vb.subhashes.p);
   std::vector<uint8_t> stuples(sjobs.size() * 128), etuples(ejobs.size() * 132);
   if (!sjobs.empty()) {
-    if (g_vb.s_jobs.ensure(sjobs.size() * sizeof(kv::kv_job)) ||
-        g_vb.s_tuples.ensure(sjobs.size() * 128))
+    if (vb.s_jobs.ensure(sjobs.size() * sizeof(kv::kv_job)) ||
+        vb.s_tuples.ensure(sjobs.size() * 128))
       return -2;
-    HIP_CHECK(hipMemcpyAsync(g_vb.s_jobs.p, sjobs.data(),
+    HIP_CHECK(hipMemcpyAsync(vb.s_jobs.p, sjobs.data(),
                              sjobs.size() * sizeof(kv::kv_job), hipMemcpyHostToDevice,
                              ctx->stream));
     hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                        dim3(((uint32_t)sjobs.size() + 255) / 256), dim3(256), 0,
-                       ctx->stream, (const uint8_t *)g_vb.blob.p,
-                       (const uint8_t *)g_vb.subhashes.p,
-                       (const kv::kv_job *)g_vb.s_jobs.p, (uint32_t)sjobs.size(),
-                       (uint8_t *)g_vb.s_tuples.p, (uint8_t *)g_vb.s_tuples.p);
-    HIP_CHECK(hipMemcpyAsync(stuples.data(), g_vb.s_tuples.p, stuples.size(),
+                       ctx->stream, (const uint8_t *)vb.blob.p,
+                       (const uint8_t *)vb.subhashes.p,
+                       (const kv::kv_job *)vb.s_jobs.p, (uint32_t)sjobs.size(),
+                       (uint8_t *)vb.s_tuples.p, (uint8_t *)vb.s_tuples.p);
+    HIP_CHECK(hipMemcpyAsync(stuples.data(), vb.s_tuples.p, stuples.size(),
                              hipMemcpyDeviceToHost, ctx->stream));
   }
   if (!ejobs.empty()) {
-    if (g_vb.e_jobs.ensure(ejobs.size() * sizeof(kv::kv_job)) ||
-        g_vb.e_tuples.ensure(ejobs.size() * 132))
+    if (vb.e_jobs.ensure(ejobs.size() * sizeof(kv::kv_job)) ||
+        vb.e_tuples.ensure(ejobs.size() * 132))
       return -2;
-    HIP_CHECK(hipMemcpyAsync(g_vb.e_jobs.p, ejobs.data(),
+    HIP_CHECK(hipMemcpyAsync(vb.e_jobs.p, ejobs.data(),
                              ejobs.size() * sizeof(kv::kv_job), hipMemcpyHostToDevice,
                              ctx->stream));
     hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                        dim3(((uint32_t)ejobs.size() + 255) / 256), dim3(256), 0,
-                       ctx->stream, (const uint8_t *)g_vb.blob.p,
-                       (const uint8_t *)g_vb.subhashes.p,
-                       (const kv::kv_job *)g_vb.e_jobs.p, (uint32_t)ejobs.size(),
-                       (uint8_t *)g_vb.e_tuples.p, (uint8_t *)g_vb.e_tuples.p);
-    HIP_CHECK(hipMemcpyAsync(etuples.data(), g_vb.e_tuples.p, etuples.size(),
+                       ctx->stream, (const uint8_t *)vb.blob.p,
+                       (const uint8_t *)vb.subhashes.p,
+                       (const kv::kv_job *)vb.e_jobs.p, (uint32_t)ejobs.size(),
+                       (uint8_t *)vb.e_tuples.p, (uint8_t *)vb.e_tuples.p);
+    HIP_CHECK(hipMemcpyAsync(etuples.data(), vb.e_tuples.p, etuples.size(),
                              hipMemcpyDeviceToHost, ctx->stream));
   }
   HIP_CHECK(hipGetLastError());
@@ -941,12 +971,13 @@ static inline int h2d_staged(kv_ctx *ctx, void *dst, const void *src, size_t len
 
 /* Enqueue the muhash element + reduce chain for txs with include[t] != 0 on
  * `stream`, fully async — the caller syncs the stream before reading outp.
- * The blob must already be resident in g_vb.blob. With no work the identity
+ * The blob must already be resident in vb.blob. With no work the identity
  * partial is written synchronously and *launched stays false. */
 static int enqueue_muhash(kv_ctx *ctx, const std::vector<HTx> &txs,
                           const uint8_t *include, uint64_t block_daa_score,
                           hipStream_t stream, bool *launched,
                           std::vector<kv::kv_elem_job> &jobs) {
+  ValidateBufs &vb = vb_of(ctx);
   if (!ctx->h_mu_partial &&
       hipHostMalloc(&ctx->h_mu_partial, 768) != hipSuccess) {
     ctx->h_mu_partial = nullptr;
@@ -981,28 +1012,28 @@ static int enqueue_muhash(kv_ctx *ctx, const std::vector<HTx> &txs,
     memcpy(outp + 384, one.l, 384);
     return 0;
   }
-  if (g_vb.elem_jobs.ensure(n_all * sizeof(kv::kv_elem_job)) ||
-      g_vb.elements.ensure(n_all * KVU_LIMBS * 8) ||
-      g_vb.partials_a.ensure(1024 * KVU_LIMBS * 8) ||
-      g_vb.partials_b.ensure(1024 * KVU_LIMBS * 8))
+  if (vb.elem_jobs.ensure(n_all * sizeof(kv::kv_elem_job)) ||
+      vb.elements.ensure(n_all * KVU_LIMBS * 8) ||
+      vb.partials_a.ensure(1024 * KVU_LIMBS * 8) ||
+      vb.partials_b.ensure(1024 * KVU_LIMBS * 8))
     return -2;
-  if (h2d_staged(ctx, g_vb.elem_jobs.p, jobs.data(),
+  if (h2d_staged(ctx, vb.elem_jobs.p, jobs.data(),
                  n_all * sizeof(kv::kv_elem_job), stream))
     return -2;
   tev_rec(ctx, 10, stream);
   hipLaunchKernelGGL(kv::kv_muhash_element_kernel,
                      dim3(((uint32_t)n_all + 255) / 256), dim3(256), 0, stream,
-                     (const uint8_t *)g_vb.blob.p,
-                     (const kv::kv_elem_job *)g_vb.elem_jobs.p, (uint32_t)n_all,
-                     (uint64_t *)g_vb.elements.p);
+                     (const uint8_t *)vb.blob.p,
+                     (const kv::kv_elem_job *)vb.elem_jobs.p, (uint32_t)n_all,
+                     (uint64_t *)vb.elements.p);
   /* reduce numerator then denominator halves down to one value each; the
    * stride schedule is pure host arithmetic, so the whole chain enqueues
    * without any device->host round trip */
   for (int half = 0; half < 2; half++) {
     size_t cnt = half == 0 ? n_num : n_den;
-    uint64_t *src = (uint64_t *)g_vb.elements.p + (half == 0 ? 0 : n_num * KVU_LIMBS);
-    uint64_t *pa = (uint64_t *)g_vb.partials_a.p;
-    uint64_t *pb = (uint64_t *)g_vb.partials_b.p;
+    uint64_t *src = (uint64_t *)vb.elements.p + (half == 0 ? 0 : n_num * KVU_LIMBS);
+    uint64_t *pa = (uint64_t *)vb.partials_a.p;
+    uint64_t *pb = (uint64_t *)vb.partials_b.p;
     if (cnt == 0) {
       memcpy(outp + half * 384, one.l, 384);
       continue;
@@ -1046,6 +1077,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     return -1;
   }
   uint64_t sigop_units = ctx->params.mass_per_sig_op * KVH_UNITS_PER_GRAM;
+  ValidateBufs &vb = vb_of(ctx);
   const bool kv_timing = getenv("KV_TIMING") != nullptr;
   auto vt_now = []() { return std::chrono::steady_clock::now(); };
   auto vt_ms = [](std::chrono::steady_clock::time_point a,
@@ -1273,40 +1305,40 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
   std::vector<uint8_t> s_gpu(ns), e_gpu(ne);
   bool blob_uploaded = false, subhash_done = false;
   if (ns + ne > 0) {
-    if (g_vb.blob.ensure(blob_len) || g_vb.subhashes.ensure((size_t)n_txs * 160))
+    if (vb.blob.ensure(blob_len) || vb.subhashes.ensure((size_t)n_txs * 160))
       return -2;
-    if (h2d_staged(ctx, g_vb.blob.p, blob, blob_len, ctx->stream)) return -2;
+    if (h2d_staged(ctx, vb.blob.p, blob, blob_len, ctx->stream)) return -2;
     HIP_CHECK(hipEventRecord(ctx->ev_blob, ctx->stream));
     blob_uploaded = true;
     tev_rec(ctx, 0);
     hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256), dim3(256),
-                       0, ctx->stream, (const uint8_t *)g_vb.blob.p, (uint32_t)n_txs,
-                       (uint8_t *)g_vb.subhashes.p);
+                       0, ctx->stream, (const uint8_t *)vb.blob.p, (uint32_t)n_txs,
+                       (uint8_t *)vb.subhashes.p);
     tev_rec(ctx, 1);
     tev_rec_pair[0] = true;
     HIP_CHECK(hipEventRecord(ctx->ev_sub, ctx->stream));
     subhash_done = true;
     if (ns) {
-      if (g_vb.s_jobs.ensure(ns * sizeof(kv::kv_job)) ||
-          g_vb.s_tuples.ensure(ns * 128) || g_vb.s_bitmap.ensure((ns + 63) / 64 * 8) ||
-          g_vb.s_status.ensure(ns))
+      if (vb.s_jobs.ensure(ns * sizeof(kv::kv_job)) ||
+          vb.s_tuples.ensure(ns * 128) || vb.s_bitmap.ensure((ns + 63) / 64 * 8) ||
+          vb.s_status.ensure(ns))
         return -2;
-      if (h2d_staged(ctx, g_vb.s_jobs.p, sjobs.data(), ns * sizeof(kv::kv_job),
+      if (h2d_staged(ctx, vb.s_jobs.p, sjobs.data(), ns * sizeof(kv::kv_job),
                      ctx->stream))
         return -2;
       tev_rec(ctx, 2);
       hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                          dim3(((uint32_t)ns + 255) / 256), dim3(256), 0, ctx->stream,
-                         (const uint8_t *)g_vb.blob.p, (const uint8_t *)g_vb.subhashes.p,
-                         (const kv::kv_job *)g_vb.s_jobs.p, (uint32_t)ns,
-                         (uint8_t *)g_vb.s_tuples.p, (uint8_t *)g_vb.s_tuples.p);
+                         (const uint8_t *)vb.blob.p, (const uint8_t *)vb.subhashes.p,
+                         (const kv::kv_job *)vb.s_jobs.p, (uint32_t)ns,
+                         (uint8_t *)vb.s_tuples.p, (uint8_t *)vb.s_tuples.p);
       tev_rec(ctx, 3);
       tev_rec_pair[1] = true;
       tev_rec(ctx, 6);
       hipLaunchKernelGGL(kv::kv_schnorr_verify_kernel, dim3(((uint32_t)ns + 255) / 256),
-                         dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.s_tuples.p,
-                         (unsigned long long)ns, (unsigned long long *)g_vb.s_bitmap.p,
-                         (uint8_t *)g_vb.s_status.p);
+                         dim3(256), 0, ctx->stream, (const uint8_t *)vb.s_tuples.p,
+                         (unsigned long long)ns, (unsigned long long *)vb.s_bitmap.p,
+                         (uint8_t *)vb.s_status.p);
       tev_rec(ctx, 7);
       tev_rec_pair[3] = true;
       ctx->last_timings.n_schnorr += ns;
@@ -1316,27 +1348,27 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
        * tiny latency-bound dispatches (a few hundred waves on a 2048-wave
        * machine) — overlapping them hides the whole ecdsa chain behind the
        * schnorr one */
-      if (g_vb.e_jobs.ensure(ne * sizeof(kv::kv_job)) ||
-          g_vb.e_tuples.ensure(ne * 132) || g_vb.e_bitmap.ensure((ne + 63) / 64 * 8) ||
-          g_vb.e_status.ensure(ne))
+      if (vb.e_jobs.ensure(ne * sizeof(kv::kv_job)) ||
+          vb.e_tuples.ensure(ne * 132) || vb.e_bitmap.ensure((ne + 63) / 64 * 8) ||
+          vb.e_status.ensure(ne))
         return -2;
       HIP_CHECK(hipStreamWaitEvent(ctx->stream2, ctx->ev_sub, 0));
-      if (h2d_staged(ctx, g_vb.e_jobs.p, ejobs.data(), ne * sizeof(kv::kv_job),
+      if (h2d_staged(ctx, vb.e_jobs.p, ejobs.data(), ne * sizeof(kv::kv_job),
                      ctx->stream2))
         return -2;
       tev_rec(ctx, 4, ctx->stream2);
       hipLaunchKernelGGL(kv::kv_sighash_assemble_kernel,
                          dim3(((uint32_t)ne + 255) / 256), dim3(256), 0, ctx->stream2,
-                         (const uint8_t *)g_vb.blob.p, (const uint8_t *)g_vb.subhashes.p,
-                         (const kv::kv_job *)g_vb.e_jobs.p, (uint32_t)ne,
-                         (uint8_t *)g_vb.e_tuples.p, (uint8_t *)g_vb.e_tuples.p);
+                         (const uint8_t *)vb.blob.p, (const uint8_t *)vb.subhashes.p,
+                         (const kv::kv_job *)vb.e_jobs.p, (uint32_t)ne,
+                         (uint8_t *)vb.e_tuples.p, (uint8_t *)vb.e_tuples.p);
       tev_rec(ctx, 5, ctx->stream2);
       tev_rec_pair[2] = true;
       tev_rec(ctx, 8, ctx->stream2);
       hipLaunchKernelGGL(kv::kv_ecdsa_verify_kernel, dim3(((uint32_t)ne + 255) / 256),
-                         dim3(256), 0, ctx->stream2, (const uint8_t *)g_vb.e_tuples.p,
-                         (unsigned long long)ne, (unsigned long long *)g_vb.e_bitmap.p,
-                         (uint8_t *)g_vb.e_status.p);
+                         dim3(256), 0, ctx->stream2, (const uint8_t *)vb.e_tuples.p,
+                         (unsigned long long)ne, (unsigned long long *)vb.e_bitmap.p,
+                         (uint8_t *)vb.e_status.p);
       tev_rec(ctx, 9, ctx->stream2);
       tev_rec_pair[4] = true;
       ctx->last_timings.n_ecdsa += ne;
@@ -1356,8 +1388,8 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
     mu_inc.resize(n_txs);
     for (int t = 0; t < n_txs; t++) mu_inc[t] = codes[t] == 0;
     if (!blob_uploaded) {
-      if (g_vb.blob.ensure(blob_len)) return -2;
-      if (h2d_staged(ctx, g_vb.blob.p, blob, blob_len, ctx->stream)) return -2;
+      if (vb.blob.ensure(blob_len)) return -2;
+      if (h2d_staged(ctx, vb.blob.p, blob, blob_len, ctx->stream)) return -2;
       HIP_CHECK(hipEventRecord(ctx->ev_blob, ctx->stream));
       blob_uploaded = true;
     }
@@ -1380,7 +1412,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
       }
       ctx->h_s_cap = nc;
     }
-    HIP_CHECK(hipMemcpyAsync(ctx->h_s_status, g_vb.s_status.p, ns,
+    HIP_CHECK(hipMemcpyAsync(ctx->h_s_status, vb.s_status.p, ns,
                              hipMemcpyDeviceToHost, ctx->stream));
   }
   if (ne) {
@@ -1395,7 +1427,7 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
       }
       ctx->h_e_cap = nc;
     }
-    HIP_CHECK(hipMemcpyAsync(ctx->h_e_status, g_vb.e_status.p, ne,
+    HIP_CHECK(hipMemcpyAsync(ctx->h_e_status, vb.e_status.p, ne,
                              hipMemcpyDeviceToHost, ctx->stream2));
   }
   auto vt2 = vt_now();
@@ -1488,79 +1520,79 @@ static int validate_block_impl(kv_ctx *ctx, const uint8_t *blob, size_t blob_len
       std::vector<uint8_t> s_st(ns_i), e_st(ne_i);
       if (ns_i + ne_i > 0) {
         if (!blob_uploaded) {
-          if (g_vb.blob.ensure(blob_len)) return -2;
-          HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len,
+          if (vb.blob.ensure(blob_len)) return -2;
+          HIP_CHECK(hipMemcpyAsync(vb.blob.p, blob, blob_len,
                                    hipMemcpyHostToDevice, ctx->stream));
           blob_uploaded = true;
         }
         if (!subhash_done) {
-          if (g_vb.subhashes.ensure((size_t)n_txs * 160)) return -2;
+          if (vb.subhashes.ensure((size_t)n_txs * 160)) return -2;
           hipLaunchKernelGGL(kv::kv_tx_subhash_kernel, dim3((n_txs + 255) / 256),
                              dim3(256), 0, ctx->stream,
-                             (const uint8_t *)g_vb.blob.p, (uint32_t)n_txs,
-                             (uint8_t *)g_vb.subhashes.p);
+                             (const uint8_t *)vb.blob.p, (uint32_t)n_txs,
+                             (uint8_t *)vb.subhashes.p);
           subhash_done = true;
         }
         if (ns_i) {
-          if (g_vb.s_tuples.ensure(st_h.size()) ||
-              g_vb.s_bitmap.ensure((ns_i + 63) / 64 * 8) ||
-              g_vb.s_status.ensure(ns_i) ||
+          if (vb.s_tuples.ensure(st_h.size()) ||
+              vb.s_bitmap.ensure((ns_i + 63) / 64 * 8) ||
+              vb.s_status.ensure(ns_i) ||
               (!mjobs_s.empty() &&
-               g_vb.s_jobs.ensure(mjobs_s.size() * sizeof(kv::kv_job))))
+               vb.s_jobs.ensure(mjobs_s.size() * sizeof(kv::kv_job))))
             return -2;
-          HIP_CHECK(hipMemcpyAsync(g_vb.s_tuples.p, st_h.data(), st_h.size(),
+          HIP_CHECK(hipMemcpyAsync(vb.s_tuples.p, st_h.data(), st_h.size(),
                                    hipMemcpyHostToDevice, ctx->stream));
           if (!mjobs_s.empty()) {
-            HIP_CHECK(hipMemcpyAsync(g_vb.s_jobs.p, mjobs_s.data(),
+            HIP_CHECK(hipMemcpyAsync(vb.s_jobs.p, mjobs_s.data(),
                                      mjobs_s.size() * sizeof(kv::kv_job),
                                      hipMemcpyHostToDevice, ctx->stream));
             hipLaunchKernelGGL(kv::kv_sighash_msg_kernel,
                                dim3(((uint32_t)mjobs_s.size() + 255) / 256),
                                dim3(256), 0, ctx->stream,
-                               (const uint8_t *)g_vb.blob.p,
-                               (const uint8_t *)g_vb.subhashes.p,
-                               (const kv::kv_job *)g_vb.s_jobs.p,
+                               (const uint8_t *)vb.blob.p,
+                               (const uint8_t *)vb.subhashes.p,
+                               (const kv::kv_job *)vb.s_jobs.p,
                                (uint32_t)mjobs_s.size(),
-                               (uint8_t *)g_vb.s_tuples.p, 128u, 96u);
+                               (uint8_t *)vb.s_tuples.p, 128u, 96u);
           }
           hipLaunchKernelGGL(kv::kv_schnorr_verify_kernel,
                              dim3(((uint32_t)ns_i + 255) / 256), dim3(256), 0,
-                             ctx->stream, (const uint8_t *)g_vb.s_tuples.p,
+                             ctx->stream, (const uint8_t *)vb.s_tuples.p,
                              (unsigned long long)ns_i,
-                             (unsigned long long *)g_vb.s_bitmap.p,
-                             (uint8_t *)g_vb.s_status.p);
-          HIP_CHECK(hipMemcpyAsync(s_st.data(), g_vb.s_status.p, ns_i,
+                             (unsigned long long *)vb.s_bitmap.p,
+                             (uint8_t *)vb.s_status.p);
+          HIP_CHECK(hipMemcpyAsync(s_st.data(), vb.s_status.p, ns_i,
                                    hipMemcpyDeviceToHost, ctx->stream));
         }
         if (ne_i) {
-          if (g_vb.e_tuples.ensure(et_h.size()) ||
-              g_vb.e_bitmap.ensure((ne_i + 63) / 64 * 8) ||
-              g_vb.e_status.ensure(ne_i) ||
+          if (vb.e_tuples.ensure(et_h.size()) ||
+              vb.e_bitmap.ensure((ne_i + 63) / 64 * 8) ||
+              vb.e_status.ensure(ne_i) ||
               (!mjobs_e.empty() &&
-               g_vb.e_jobs.ensure(mjobs_e.size() * sizeof(kv::kv_job))))
+               vb.e_jobs.ensure(mjobs_e.size() * sizeof(kv::kv_job))))
             return -2;
-          HIP_CHECK(hipMemcpyAsync(g_vb.e_tuples.p, et_h.data(), et_h.size(),
+          HIP_CHECK(hipMemcpyAsync(vb.e_tuples.p, et_h.data(), et_h.size(),
                                    hipMemcpyHostToDevice, ctx->stream));
           if (!mjobs_e.empty()) {
-            HIP_CHECK(hipMemcpyAsync(g_vb.e_jobs.p, mjobs_e.data(),
+            HIP_CHECK(hipMemcpyAsync(vb.e_jobs.p, mjobs_e.data(),
                                      mjobs_e.size() * sizeof(kv::kv_job),
                                      hipMemcpyHostToDevice, ctx->stream));
             hipLaunchKernelGGL(kv::kv_sighash_msg_kernel,
                                dim3(((uint32_t)mjobs_e.size() + 255) / 256),
                                dim3(256), 0, ctx->stream,
-                               (const uint8_t *)g_vb.blob.p,
-                               (const uint8_t *)g_vb.subhashes.p,
-                               (const kv::kv_job *)g_vb.e_jobs.p,
+                               (const uint8_t *)vb.blob.p,
+                               (const uint8_t *)vb.subhashes.p,
+                               (const kv::kv_job *)vb.e_jobs.p,
                                (uint32_t)mjobs_e.size(),
-                               (uint8_t *)g_vb.e_tuples.p, 132u, 97u);
+                               (uint8_t *)vb.e_tuples.p, 132u, 97u);
           }
           hipLaunchKernelGGL(kv::kv_ecdsa_verify_kernel,
                              dim3(((uint32_t)ne_i + 255) / 256), dim3(256), 0,
-                             ctx->stream, (const uint8_t *)g_vb.e_tuples.p,
+                             ctx->stream, (const uint8_t *)vb.e_tuples.p,
                              (unsigned long long)ne_i,
-                             (unsigned long long *)g_vb.e_bitmap.p,
-                             (uint8_t *)g_vb.e_status.p);
-          HIP_CHECK(hipMemcpyAsync(e_st.data(), g_vb.e_status.p, ne_i,
+                             (unsigned long long *)vb.e_bitmap.p,
+                             (uint8_t *)vb.e_status.p);
+          HIP_CHECK(hipMemcpyAsync(e_st.data(), vb.e_status.p, ne_i,
                                    hipMemcpyDeviceToHost, ctx->stream));
         }
         HIP_CHECK(hipGetLastError());
@@ -2095,6 +2127,7 @@ extern "C" int kv_block_body_check(kv_ctx *ctx, const uint8_t *blob,
   std::lock_guard<std::mutex> lk(ctx->mu);
   (void)hipGetLastError(); /* clear any stale per-thread error so the
       launch-config checks below only see THIS call's launches */
+  ValidateBufs &vb = vb_of(ctx);
   vector<HTx> txs;
   int n_txs = parse_blob_host(blob, blob_len, txs);
   if (n_txs < 0) {
@@ -2104,15 +2137,15 @@ extern "C" int kv_block_body_check(kv_ctx *ctx, const uint8_t *blob,
   /* leaf hashes on device */
   std::vector<uint8_t> hashes((size_t)n_txs * 32);
   if (n_txs > 0) {
-    if (g_vb.blob.ensure(blob_len) || g_vb.tx_hashes.ensure((size_t)n_txs * 32))
+    if (vb.blob.ensure(blob_len) || vb.tx_hashes.ensure((size_t)n_txs * 32))
       return -2;
-    HIP_CHECK(hipMemcpyAsync(g_vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
+    HIP_CHECK(hipMemcpyAsync(vb.blob.p, blob, blob_len, hipMemcpyHostToDevice,
                              ctx->stream));
     hipLaunchKernelGGL(kv::kv_tx_hash_kernel, dim3((n_txs + 255) / 256),
-                       dim3(256), 0, ctx->stream, (const uint8_t *)g_vb.blob.p,
-                       (uint32_t)n_txs, (uint8_t *)g_vb.tx_hashes.p);
+                       dim3(256), 0, ctx->stream, (const uint8_t *)vb.blob.p,
+                       (uint32_t)n_txs, (uint8_t *)vb.tx_hashes.p);
     HIP_CHECK(hipGetLastError());
-    HIP_CHECK(hipMemcpyAsync(hashes.data(), g_vb.tx_hashes.p,
+    HIP_CHECK(hipMemcpyAsync(hashes.data(), vb.tx_hashes.p,
                              (size_t)n_txs * 32, hipMemcpyDeviceToHost,
                              ctx->stream));
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
