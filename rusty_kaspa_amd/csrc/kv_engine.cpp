@@ -123,6 +123,7 @@ struct kv_ctx {
   size_t d_gjobs_cap = 0;
   void *d_gout = nullptr;
   size_t d_gout_cap = 0;
+  int *d_utxo_fail = nullptr;
   uint8_t *d_op_in = nullptr;
   size_t d_op_cap = 0;
   uint8_t *d_val_in = nullptr;
@@ -204,6 +205,7 @@ extern "C" void kv_destroy(kv_ctx *ctx) {
   if (ctx->tev_init)
     for (int i = 0; i < 12; i++) (void)hipEventDestroy(ctx->tev[i]);
   if (ctx->d_utxo) (void)hipFree(ctx->d_utxo);
+  if (ctx->d_utxo_fail) (void)hipFree(ctx->d_utxo_fail);
   if (ctx->d_op_in) (void)hipFree(ctx->d_op_in);
   if (ctx->d_val_in) (void)hipFree(ctx->d_val_in);
   if (ctx->d_ent_out) (void)hipFree(ctx->d_ent_out);
@@ -1765,8 +1767,12 @@ static int utxo_upsert_nolock(kv_ctx *ctx, const uint8_t *outpoints,
   }
   int rc = utxo_stage(ctx, outpoints, entries64, n);
   if (rc) return rc;
-  static int *d_fail = nullptr;
-  if (!d_fail) HIP_CHECK(hipMalloc(&d_fail, 4));
+  int *d_fail = ctx->d_utxo_fail; /* per-ctx: a shared flag would race when
+                                     two contexts upsert concurrently */
+  if (!d_fail) {
+    HIP_CHECK(hipMalloc(&d_fail, 4));
+    ctx->d_utxo_fail = d_fail;
+  }
   HIP_CHECK(hipMemsetAsync(d_fail, 0, 4, ctx->stream));
   hipLaunchKernelGGL(kv::kv_utxo_upsert_kernel, dim3(((uint32_t)n + 255) / 256),
                      dim3(256), 0, ctx->stream, ctx->d_utxo, ctx->utxo_cap - 1,
