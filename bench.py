@@ -65,12 +65,13 @@ ALG_OPS_PER_VERIFY = ALG_FE_MULS_PER_VERIFY * ALG_OPS_PER_FE_MUL  # 524,450
 # gfx950 VALU issue peak: 256 CU x 4 SIMD x 32 lanes x 2.4 GHz = 78.6 T u32/s
 VALU_PEAK_TOPS = 78.6
 # Memory-side traffic per verify, measured by rocprofv3 --pmc FETCH_SIZE /
-# WRITE_SIZE (separate passes) on 262144-tuple staged dispatches of the FINAL
-# kernel (G comb + pair-window frames): (9.226e6 + 5.536e6) KiB / 262144 =
-# 57,664 B/verify (fetch 36.0KB + write 21.6KB — scratch table reads + the
-# per-frame accumulator spill; vs 128B of algorithmic input). Provenance:
-# profiles/r02c_final_kernel_pmc.json (pre-fusion: 119,296).
-TRAFFIC_BYTES_PER_VERIFY = 57664
+# WRITE_SIZE (separate passes) on 262144-tuple staged dispatches of the
+# SHIPPING kernel (signed P table + G comb + pair-window frames):
+# (5.684e6 + 4.668e6) KiB / 262144 = 40,438 B/verify (fetch 22.2KB + write
+# 18.2KB — scratch table reads + the per-frame accumulator spill; vs 128B of
+# algorithmic input; r1 baseline 119,296). Provenance:
+# profiles/r02e_shipping_kernel_pmc.json.
+TRAFFIC_BYTES_PER_VERIFY = 40438
 
 
 def log(msg):
