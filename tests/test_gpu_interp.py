@@ -302,3 +302,33 @@ def test_forty_sites_forty_rounds(oracle, engine):
     txs[0] = dict(txs[0], inputs=ins)
     ocodes, ecodes = validate_both(oracle, engine, txs)
     assert ecodes == ocodes and ocodes[0] != 0
+
+
+def test_p2sh_noncanonical_redeem_with_sig(oracle, engine):
+    """A P2SH spend whose redeem script is NOT the canonical multisig template
+    (checksig + NOP) — the classify fast path rejects it and the general
+    interpreter runs the full [sig, spk, redeem] sequence with the signature
+    resolved on GPU."""
+    key, pk = keypair(oracle, 33)
+    redeem = push(pk) + b"\xac\x61"  # <pk> CHECKSIG NOP — non-template
+    # spk = OpBlake2b <32B hash> OpEqual (0xaa 0x20 h 0x87)
+    import hashlib
+    h = hashlib.blake2b(redeem, digest_size=32).digest()
+    spk = b"\xaa\x20" + h + b"\x87"
+    placeholder = push(bytes(65)) + push(redeem)
+    txs = [spend_tx(oracle, placeholder, spk)]
+    blob = B.build_blob(txs)
+    msg = (ctypes.c_uint8 * 32)()
+    assert oracle.ok_sighash(blob, len(blob), 0, 0, 1, 0, msg) == 0
+    sig = (ctypes.c_uint8 * 64)()
+    assert oracle.ok_schnorr_sign(key, bytes(msg), None, sig) == 1
+    ins = list(txs[0]["inputs"])
+    ins[0] = dict(ins[0], sig_script=push(bytes(sig) + b"\x01") + push(redeem))
+    txs[0] = dict(txs[0], inputs=ins)
+    ocodes, ecodes = validate_both(oracle, engine, txs)
+    assert ecodes == ocodes == [0], (ocodes, ecodes)
+    # wrong redeem hash rejects identically (EvalFalse from the spk phase)
+    bad_spk = b"\xaa\x20" + bytes(32) + b"\x87"
+    txs2 = [spend_tx(oracle, ins[0]["sig_script"], bad_spk, prev_seed=1)]
+    ocodes, ecodes = validate_both(oracle, engine, txs2)
+    assert ecodes == ocodes and ocodes[0] != 0

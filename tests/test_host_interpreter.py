@@ -189,7 +189,7 @@ def test_sig_family_structured_fuzz(oracle, shim, verifier):
         assert len(data) <= 0x4b
         return bytes([len(data)]) + data if data else b"\x00"
 
-    for trial in range(1500):
+    for trial in range(4000):
         mode = rng.randrange(6)
         key = bytearray(rng.randrange(256) for _ in range(32))
         sig64 = bytearray(rng.randrange(256) for _ in range(64))
